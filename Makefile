@@ -1,0 +1,37 @@
+# Top-level build: GPU apply-path library (libgra.so, gfx950) + CPU oracle.
+# Built IN-TREE so the .so files travel with the gpurun snapshot.
+HIPCC ?= hipcc
+ARCH ?= gfx950
+HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC -Wall
+
+CSRC := rocksplicator_amd/csrc
+OBJS := build/engine.o build/host_store.o build/builder.o build/gen.o
+
+all: rocksplicator_amd/libgra.so oracle/libwb_oracle.so
+
+build:
+	mkdir -p build
+
+build/engine.o: $(CSRC)/engine.hip $(CSRC)/wb_format.h $(CSRC)/host_store.h include/rocksplicator_gpu.h | build
+	$(HIPCC) $(HIPFLAGS) -x hip -c $< -o $@
+
+build/host_store.o: $(CSRC)/host_store.cpp $(CSRC)/host_store.h $(CSRC)/wb_format.h | build
+	$(HIPCC) $(HIPFLAGS) -c $< -o $@
+
+build/builder.o: $(CSRC)/builder.cpp $(CSRC)/wb_format.h include/rocksplicator_gpu.h | build
+	$(HIPCC) $(HIPFLAGS) -c $< -o $@
+
+build/gen.o: $(CSRC)/gen.cpp $(CSRC)/wb_format.h include/rocksplicator_gpu.h | build
+	$(HIPCC) $(HIPFLAGS) -c $< -o $@
+
+rocksplicator_amd/libgra.so: $(OBJS)
+	$(HIPCC) $(HIPFLAGS) -shared $(OBJS) -o $@
+
+oracle/libwb_oracle.so: oracle/wb_oracle.c oracle/wb_oracle.h
+	$(MAKE) -C oracle
+
+clean:
+	rm -rf build rocksplicator_amd/libgra.so
+	$(MAKE) -C oracle clean
+
+.PHONY: all clean

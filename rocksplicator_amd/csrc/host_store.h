@@ -1,0 +1,59 @@
+/* host_store.h — host-side run registry standing behind the DbWrapper seam.
+ *
+ * MI355X-first design: the follower "memtable" is a sequence of GPU-produced
+ * sorted-by-seq runs resident in device HBM (288 GB/GPU); the host keeps only
+ * run descriptors and lazily fetches run bytes for parity probes (gra_get).
+ * Leader-side writes (WriteToLeader, rocksdb_wrapper.cpp:5-8) produce
+ * host-built runs in the identical format.
+ *
+ * Get semantics must equal the oracle's (tests/test_oracle.py): walk runs
+ * newest→oldest, entries newest→oldest; Put = base, Delete/SingleDelete =
+ * tombstone, Merge = operand collected then folded oldest→newest with the
+ * engine merge operator, RangeDeletion = tombstone for covered keys below
+ * its seq.
+ */
+#pragma once
+#include <cstdint>
+#include <cstring>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "wb_format.h"
+
+namespace gra {
+
+struct Run {
+  uint64_t base_seq = 0, last_seq = 0;
+  uint32_t n_entries = 0;
+  uint32_t payload_bytes = 0;
+  /* device-resident location (absolute offsets into the store arena);
+   * UINT64_MAX for host-origin (leader-write) runs */
+  uint64_t hdr_cur = UINT64_MAX, payload_cur = UINT64_MAX;
+  uint32_t pay_rel_base = 0; /* device runs: kv_off rebase at fetch time */
+  /* host copies — always set for host runs, lazily fetched for device runs */
+  std::vector<uint8_t> hdrs;     /* n_entries × wb::RecHdr */
+  std::vector<uint8_t> payload;
+  bool resident() const { return !hdrs.empty() || n_entries == 0; }
+};
+
+struct ShardState {
+  mutable std::mutex mu;
+  uint64_t durable_seq = 0;   /* last seq applied & synced on device */
+  uint64_t next_seq = 1;      /* next seq to assign at submission */
+  bool poisoned = false;      /* corrupt batch seen; next HRR returns false */
+  std::vector<std::shared_ptr<Run>> runs; /* oldest .. newest */
+};
+
+/* Get over a run list (newest last). merge_op: 0 concat, 1 u64add.
+ * Returns 0 found, 1 not found. Found value appended to out. */
+int run_get(const std::vector<std::shared_ptr<Run>> &runs, const void *key,
+            size_t klen, int merge_op, std::string *out);
+
+/* Host apply of one rep blob (leader write path): decodes with wb::walk and
+ * builds a Run in the same format the GPU emits. Returns false on corrupt
+ * rep. base_seq = first seq to assign. */
+bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out);
+
+} /* namespace gra */
