@@ -363,6 +363,7 @@ struct GraEngine {
   /* tick scratch */
   uint32_t max_upd;          /* max updates per tick */
   uint32_t task_cap;
+  uint32_t group_cap;        /* max contiguous shard-groups per tick */
   wb::WalkTotals *d_totals = nullptr;
   uint2 *d_partial = nullptr, *d_bsums = nullptr, *d_scan = nullptr;
   CopyTask *d_tasks = nullptr;
@@ -438,6 +439,7 @@ int GraEngine::init(const GraEngineOpts &o) {
   if (opts.max_wb_records == 0) opts.max_wb_records = 1024;
   max_upd = 1u << 20;
   task_cap = 4u << 20;
+  group_cap = opts.nshards + 1 > 65536 ? opts.nshards + 1 : 65536;
   int ndev = 0;
   hipError_t de = hipGetDeviceCount(&ndev);
   if (de != hipSuccess || ndev == 0) {
@@ -456,8 +458,8 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMalloc(&d_scan, ((size_t)max_upd + 1) * sizeof(uint2)));
   HIP_TRY(hipMalloc(&d_tasks, (size_t)task_cap * sizeof(CopyTask)));
   HIP_TRY(hipMalloc(&d_place, sizeof(TickPlace)));
-  HIP_TRY(hipMalloc(&d_groups, ((size_t)opts.nshards + 1) * sizeof(GroupDesc)));
-  HIP_TRY(hipMalloc(&d_rundescs, ((size_t)opts.nshards + 1) * sizeof(DevRunDesc)));
+  HIP_TRY(hipMalloc(&d_groups, (size_t)group_cap * sizeof(GroupDesc)));
+  HIP_TRY(hipMalloc(&d_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
   HIP_TRY(hipMalloc(&d_err, 4));
   HIP_TRY(hipMalloc(&d_stage_blobs, opts.staging_bytes + 16));
   HIP_TRY(hipMalloc(&d_stage_descs, (size_t)max_upd * sizeof(UpdDesc)));
@@ -468,8 +470,8 @@ int GraEngine::init(const GraEngineOpts &o) {
   }
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
-    HIP_TRY(hipHostMalloc(&s.h_groups, ((size_t)opts.nshards + 1) * sizeof(GroupDesc)));
-    HIP_TRY(hipHostMalloc(&s.h_rundescs, ((size_t)opts.nshards + 1) * sizeof(DevRunDesc)));
+    HIP_TRY(hipHostMalloc(&s.h_groups, (size_t)group_cap * sizeof(GroupDesc)));
+    HIP_TRY(hipHostMalloc(&s.h_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
     HIP_TRY(hipHostMalloc(&s.h_recn, (size_t)max_upd * 2));
     HIP_TRY(hipHostMalloc(&s.h_err, 4));
     HIP_TRY(hipHostMalloc(&s.h_descs, (size_t)max_upd * sizeof(UpdDesc)));
@@ -528,6 +530,10 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
+    return GRA_ERR;
+  }
+  if (groups.size() > group_cap) {
+    g_err = "tick exceeds max shard-groups per tick";
     return GRA_ERR;
   }
   int si = free_slot();
@@ -933,17 +939,14 @@ static int build_groups(GraReplay *r, uint64_t first, uint64_t n,
   uint64_t bb = 0;
   hu.resize(n);
   uint32_t cur_shard = UINT32_MAX;
-  std::vector<uint8_t> seen(r->e->opts.nshards, 0);
+  /* each contiguous same-shard range becomes one run; a shard may appear in
+   * several ranges per tick (runs are ingested in tick order, so per-shard
+   * seq order is preserved) */
   for (uint64_t i = 0; i < n; i++) {
     const UpdDesc &d = r->descs[first + i];
     bb += d.len;
     hu[i] = {d.shard, r->counts[first + i], d.base_seq};
     if (d.shard != cur_shard) {
-      if (seen[d.shard]) {
-        g_err = "replay window not shard-grouped";
-        return GRA_ERR;
-      }
-      seen[d.shard] = 1;
       groups.push_back({d.shard, (uint32_t)i, 1, 0});
       cur_shard = d.shard;
     } else {

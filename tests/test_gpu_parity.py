@@ -1,0 +1,237 @@
+"""GPU parity tests — the follower apply path vs the CPU oracle on identical
+update streams. Bit-exact bar: latest seq per shard and per-key Get equality
+(the reference's own verification pattern, rocksdb_replicator_test.cpp:
+162-207 and rocksdb_assumption_test.cpp:329-432).
+
+All tests here require a real MI355X (@pytest.mark.gpu) and exercise the
+HIP pipeline through the C-ABI — there is no CPU fallback to hide behind.
+"""
+import ctypes as C
+import random
+
+import pytest
+
+import oracle_ffi
+import rocksplicator_amd as ra
+from pywb import PyBatch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def olib():
+    return oracle_ffi.load()
+
+
+def oracle_apply_stream(olib, nshards, raw, descs, n, merge_op=0):
+    st = oracle_ffi.Store(olib, nshards, merge_op=merge_op)
+    for i in range(n):
+        d = descs[i]
+        assert st.apply(d.shard, raw[d.off:d.off + d.len], d.ts)
+    return st
+
+
+def collect_keys(olib, raw, descs, n, per_shard_cap=64):
+    """shard -> list of distinct full keys seen (for Get probes)."""
+    keys = {}
+    for i in range(n):
+        d = descs[i]
+        blob = raw[d.off:d.off + d.len]
+        _, _, recs = oracle_ffi.decode(olib, blob)
+        for r in recs:
+            if not r.consumes_seq or r.type == 0x0F:
+                continue
+            k = blob[r.key_off:r.key_off + r.key_len]
+            keys.setdefault(d.shard, [])
+            if len(keys[d.shard]) < per_shard_cap and k not in keys[d.shard]:
+                keys[d.shard].append(k)
+    return keys
+
+
+def check_parity(engine, olib_store, keys, shards):
+    for s in shards:
+        db = engine.open(s)
+        assert db.latest_seq() == olib_store.latest_seq(s), f"shard {s} seq"
+        for k in keys.get(s, []):
+            assert db.get(k) == olib_store.get(s, k), f"shard {s} key {k.hex()}"
+        db.close()
+
+
+# ---------------- streaming ingest (HandleReplicateResponse) ----------------
+
+def test_stream_small_batches(olib):
+    e = ra.Engine(nshards=4)
+    dbs = [e.open(s) for s in range(4)]
+    ost = oracle_ffi.Store(olib, 4)
+    rng = random.Random(1)
+    blobs = []
+    for i in range(200):
+        s = rng.randrange(4)
+        b = PyBatch()
+        k = f"key{rng.randrange(50)}".encode()
+        if rng.random() < 0.7:
+            b.put(k, rng.randbytes(rng.randrange(1, 200)))
+        else:
+            b.delete(k)
+        blobs.append((s, b.data()))
+    for s, rep in blobs:
+        assert dbs[s].handle_replicate_response(rep, ts=123)
+        assert ost.apply(s, rep, 123)
+    e.flush()
+    for s in range(4):
+        assert dbs[s].latest_seq() == ost.latest_seq(s)
+        for i in range(50):
+            k = f"key{i}".encode()
+            assert dbs[s].get(k) == ost.get(s, k), (s, k)
+    e.close()
+
+
+def test_stream_multi_record_batches(olib):
+    e = ra.Engine(nshards=2)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 2)
+    # the assumption-test batch shape (del, put, put, merge) + logdata trailer
+    rep = (PyBatch().delete(b"key1").put(b"key2", b"value2")
+           .put(b"key2", b"value2").merge(b"key1", b"value1")
+           .log_data(b"12345678").data())
+    assert db.handle_replicate_response(rep)
+    assert ost.apply(0, rep)
+    rep2 = PyBatch().put(b"key1", b"v1").data()
+    assert db.handle_replicate_response(rep2)
+    assert ost.apply(0, rep2)
+    e.flush()
+    assert db.latest_seq() == ost.latest_seq(0) == 5
+    for k in (b"key1", b"key2", b"nope"):
+        assert db.get(k) == ost.get(0, k)
+    e.close()
+
+
+def test_stream_edge_cases(olib):
+    e = ra.Engine(nshards=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1)
+    cases = [
+        PyBatch().log_data(b"only-log-data").data(),       # count 0
+        PyBatch().put(b"", b"empty-key-value").data(),     # empty key
+        PyBatch().put(b"ek", b"").data(),                  # empty value
+        PyBatch().put(b"big", bytes(range(256)) * 256).data(),  # 64 KB value
+        PyBatch().delete_range(b"a", b"c").data(),
+        PyBatch().cf_put(3, b"cfk", b"cfv").data(),        # cf-prefixed
+        PyBatch().single_delete(b"ek").data(),
+    ]
+    for rep in cases:
+        assert db.handle_replicate_response(rep)
+        assert ost.apply(0, rep)
+    e.flush()
+    assert db.latest_seq() == ost.latest_seq(0)
+    for k in (b"", b"ek", b"big", b"a", b"b", b"cfk"):
+        assert db.get(k) == ost.get(0, k), k
+    e.close()
+
+
+def test_corrupt_batch_poisons_shard(olib):
+    e = ra.Engine(nshards=2)
+    db = e.open(0)
+    good = PyBatch().put(b"k", b"v").data()
+    assert db.handle_replicate_response(good)
+    e.flush()
+    assert db.latest_seq() == 1
+    bad = bytearray(PyBatch().put(b"x", b"y").data())
+    bad[8] = 9  # count mismatch -> corruption on GPU validation
+    assert db.handle_replicate_response(bytes(bad))  # accepted (async)
+    e.flush()
+    # poisoned: next call fails once (reference delayed-re-pull cadence),
+    # latest_seq reports the durable seq
+    assert db.latest_seq() == 1
+    assert not db.handle_replicate_response(good)
+    # after the failure signal, the shard recovers
+    assert db.handle_replicate_response(good)
+    e.flush()
+    assert db.latest_seq() == 2
+    assert db.get(b"k") == b"v"
+    e.close()
+
+
+# ---------------- replay path (blobs resident in HBM) ----------------
+
+@pytest.mark.parametrize("kind,merge_op,nupd,nshards,vlen", [
+    (0, 0, 20000, 64, 128),   # config #2 shape (scaled down)
+    (1, 0, 20000, 256, 256),  # zipf keys
+    (2, 1, 20000, 64, 64),    # mixed put/delete/merge with u64add
+])
+def test_replay_parity(olib, kind, merge_op, nupd, nshards, vlen):
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=vlen, kind=kind,
+                                       key_space=1 << 16, seed=99 + kind)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd, merge_op=merge_op)
+    e = ra.Engine(nshards=nshards, merge_op=merge_op)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    # several ticks, windows must be shard-grouped (generator guarantees it
+    # only at 50-update boundaries)
+    tick = 5000
+    for first in range(0, nupd, tick):
+        rep.tick(first, min(tick, nupd - first))
+    rep.sync()
+    keys = collect_keys(olib, raw, descs, min(nupd, 4000))
+    check_parity(e, ost, keys, range(0, nshards, max(1, nshards // 16)))
+    s = e.stats()
+    assert s.updates == nupd
+    assert s.records == nupd  # 1 record per update in generator streams
+    e.close()
+
+
+def test_replay_h2d_parity(olib):
+    nshards, nupd = 32, 10000
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=128, seed=5)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd)
+    e = ra.Engine(nshards=nshards)
+    # pinned arena for PCIe-inclusive path
+    pin = e.pin_alloc(used)
+    C.memmove(pin, arena, used)
+    rep = e.upload(pin, used, descs, nupd)
+    for first in range(0, nupd, 2500):
+        rep.tick_h2d(first, 2500)
+    rep.sync()
+    keys = collect_keys(olib, raw, descs, 4000)
+    check_parity(e, ost, keys, range(0, nshards, 4))
+    st = e.stats()
+    assert st.h2d_ms > 0
+    e.close()
+
+
+def test_leader_write_then_get_on_gpu_box(olib):
+    """Host-side leader path (WriteToLeader semantics) — same run format."""
+    e = ra.Engine(nshards=1, merge_op=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1, merge_op=1)
+    one = (1).to_bytes(8, "little")
+    for i in range(10):
+        rep = PyBatch().merge(b"ctr", one).data()
+        seq = db.write_leader(rep)
+        assert ost.apply(0, rep)
+        assert seq == ost.latest_seq(0)
+    assert db.get(b"ctr") == ost.get(0, b"ctr")
+    assert int.from_bytes(db.get(b"ctr"), "little") == 10
+    e.close()
+
+
+def test_mixed_leader_and_replicated(olib):
+    """Leader-written host runs and GPU-applied device runs interleave in one
+    shard's run list; Get must merge across both."""
+    e = ra.Engine(nshards=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1)
+    db.write_leader(PyBatch().put(b"a", b"host1").data())
+    ost.apply(0, PyBatch().put(b"a", b"host1").data())
+    assert db.handle_replicate_response(PyBatch().put(b"a", b"gpu1").put(b"b", b"gpu2").data())
+    ost.apply(0, PyBatch().put(b"a", b"gpu1").put(b"b", b"gpu2").data())
+    e.flush()
+    db.write_leader(PyBatch().delete(b"b").data())
+    ost.apply(0, PyBatch().delete(b"b").data())
+    assert db.latest_seq() == ost.latest_seq(0) == 4
+    for k in (b"a", b"b"):
+        assert db.get(k) == ost.get(0, k)
+    e.close()
