@@ -82,20 +82,22 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s):
             olib.orc_store_destroy(st)
         return secs
 
-    probe_n = min(n, 50000)
-    secs = run(probe_n, ncores)
-    rate = probe_n / secs if secs > 0 else 1.0
-    sample_n = min(n, max(probe_n, int(rate * target_s)))
-    if sample_n > probe_n:
+    # bounded sample: repeat full-stream applies until ~target_s of CPU work
+    total_n, total_s, reps = 0, 0.0, 0
+    sample_n = min(n, 2_000_000)
+    while total_s < target_s and reps < 64:
         secs = run(sample_n, ncores)
-        rate = sample_n / secs
+        total_n += sample_n
+        total_s += secs
+        reps += 1
+    rate = total_n / total_s if total_s > 0 else 0.0
     return {
         "value": rate,
         "unit": "updates/s",
         "cores": ncores,
         "kind": "port",
-        "sample": f"{sample_n} updates of the same synthetic stream, "
-                  f"{ncores} threads, {secs:.2f}s",
+        "sample": f"{reps}x{sample_n} updates of the same synthetic stream, "
+                  f"{ncores} threads, {total_s:.2f}s total",
     }
 
 

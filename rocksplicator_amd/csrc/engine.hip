@@ -32,6 +32,7 @@
 #include <cstdio>
 #include <cstring>
 #include <deque>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -325,18 +326,11 @@ struct Stats {
 constexpr int kSlots = 8;
 constexpr int kEventsPerTick = 8;
 
-struct HostUpd { /* host bookkeeping per update in a tick */
-  uint32_t shard;
-  uint32_t count;
-  uint64_t base_seq;
-};
-
 struct TickRec {
   int slot = -1;
   uint32_t n = 0, ngroups = 0;
   uint64_t blob_bytes = 0;
   bool h2d_timed = false;
-  std::vector<HostUpd> upds; /* per-update bookkeeping for error recovery */
   hipEvent_t ev[kEventsPerTick];
 };
 
@@ -391,8 +385,7 @@ struct GraEngine {
   int init(const GraEngineOpts &o);
   ~GraEngine();
   int enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw, uint32_t n,
-                   const std::vector<GroupDesc> &groups,
-                   std::vector<HostUpd> &&upds, uint64_t blob_bytes,
+                   const std::vector<GroupDesc> &groups, uint64_t blob_bytes,
                    bool time_h2d, const void *h2d_src = nullptr,
                    size_t h2d_bytes = 0, uint8_t *d_h2d_dst = nullptr,
                    const UpdDesc *h_descs_src = nullptr);
@@ -410,6 +403,11 @@ struct GraDb {
   uint32_t shard;
 };
 
+struct TickPlan {
+  std::vector<GroupDesc> groups;
+  uint64_t blob_bytes = 0;
+};
+
 struct GraReplay {
   GraEngine *e = nullptr;
   uint8_t *d_blobs = nullptr;
@@ -418,6 +416,7 @@ struct GraReplay {
   std::vector<uint16_t> counts;
   size_t arena_bytes = 0;
   const uint8_t *h_arena = nullptr; /* for tick_h2d (must stay alive) */
+  std::map<std::pair<uint64_t, uint64_t>, TickPlan> plans; /* window cache */
 };
 
 hipEvent_t GraEngine::get_event() {
@@ -523,10 +522,9 @@ int GraEngine::free_slot() {
 
 int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                             uint32_t n, const std::vector<GroupDesc> &groups,
-                            std::vector<HostUpd> &&upds, uint64_t blob_bytes,
-                            bool time_h2d, const void *h2d_src,
-                            size_t h2d_bytes, uint8_t *d_h2d_dst,
-                            const UpdDesc *h_descs_src) {
+                            uint64_t blob_bytes, bool time_h2d,
+                            const void *h2d_src, size_t h2d_bytes,
+                            uint8_t *d_h2d_dst, const UpdDesc *h_descs_src) {
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
@@ -549,7 +547,6 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   t.ngroups = ngroups;
   t.blob_bytes = blob_bytes;
   t.h2d_timed = time_h2d;
-  t.upds = std::move(upds);
   for (int i = 0; i < kEventsPerTick; i++) t.ev[i] = get_event();
 
   HIP_TRY(hipMemsetAsync(d_err, 0, 4, stream));
@@ -650,6 +647,20 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     sl.busy = false;
     return GRA_OK;
   }
+  if (opts.store_ring) {
+    /* throughput store: runs are recycled by the ring; keep only seq/stats
+     * bookkeeping (Get over recycled regions is undefined by contract) */
+    for (uint32_t g = 0; g < t.ngroups; g++) {
+      DevRunDesc &rd = sl.h_rundescs[g];
+      ShardState &ss = shards[rd.shard];
+      if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
+      stats.records += rd.n_entries;
+      stats.payload_bytes += rd.payload_bytes;
+    }
+    for (int i = 0; i < kEventsPerTick; i++) put_event(t.ev[i]);
+    sl.busy = false;
+    return GRA_OK;
+  }
   for (uint32_t g = 0; g < t.ngroups; g++) {
     DevRunDesc &rd = sl.h_rundescs[g];
     ShardState &ss = shards[rd.shard];
@@ -691,29 +702,25 @@ int GraEngine::stream_tick_locked() {
   for (auto &d : stage_descs) cnt[d.shard + 1]++;
   for (uint32_t s = 0; s < opts.nshards; s++) cnt[s + 1] += cnt[s];
   std::vector<UpdDesc> ud(n);
-  std::vector<HostUpd> hu(n);
   std::vector<GroupDesc> groups;
   uint64_t blob_bytes = 0;
   {
     std::vector<uint32_t> pos = cnt;
     for (auto &d : stage_descs) {
       uint32_t i = pos[d.shard]++;
-      /* base_seq was assigned at submit; stash via ts? no — recompute: we
-       * stored base_seq in desc.ts field for staging-path descs */
       ud[i].off = d.off;
       ud[i].len = d.len;
       ud[i].shard = d.shard;
-      ud[i].base_seq = (uint64_t)d.ts; /* staged base_seq */
-      hu[i] = {d.shard, 0, ud[i].base_seq};
+      ud[i].base_seq = (uint64_t)d.ts; /* base_seq stashed at submit */
       blob_bytes += d.len;
     }
     for (uint32_t s = 0; s < opts.nshards; s++)
       if (cnt[s + 1] > cnt[s])
         groups.push_back({s, cnt[s], cnt[s + 1] - cnt[s], 0});
   }
-  int rc = enqueue_tick(d_stage_blobs, d_stage_descs, n, groups, std::move(hu),
-                        blob_bytes, true, h_stage[cur], stage_fill,
-                        d_stage_blobs, ud.data());
+  int rc = enqueue_tick(d_stage_blobs, d_stage_descs, n, groups, blob_bytes,
+                        true, h_stage[cur], stage_fill, d_stage_blobs,
+                        ud.data());
   if (rc != GRA_OK) return rc;
   /* mark this pinned buffer reusable once the H2D completed */
   HIP_TRY(hipEventRecord(stage_free[cur], stream));
@@ -933,11 +940,12 @@ void gra_replay_destroy(GraReplay *r) {
   delete r;
 }
 
-static int build_groups(GraReplay *r, uint64_t first, uint64_t n,
-                        std::vector<GroupDesc> &groups,
-                        std::vector<HostUpd> &hu, uint64_t *blob_bytes) {
+static const TickPlan &plan_for(GraReplay *r, uint64_t first, uint64_t n) {
+  auto key = std::make_pair(first, n);
+  auto it = r->plans.find(key);
+  if (it != r->plans.end()) return it->second;
+  TickPlan plan;
   uint64_t bb = 0;
-  hu.resize(n);
   uint32_t cur_shard = UINT32_MAX;
   /* each contiguous same-shard range becomes one run; a shard may appear in
    * several ranges per tick (runs are ingested in tick order, so per-shard
@@ -945,16 +953,15 @@ static int build_groups(GraReplay *r, uint64_t first, uint64_t n,
   for (uint64_t i = 0; i < n; i++) {
     const UpdDesc &d = r->descs[first + i];
     bb += d.len;
-    hu[i] = {d.shard, r->counts[first + i], d.base_seq};
     if (d.shard != cur_shard) {
-      groups.push_back({d.shard, (uint32_t)i, 1, 0});
+      plan.groups.push_back({d.shard, (uint32_t)i, 1, 0});
       cur_shard = d.shard;
     } else {
-      groups.back().n_upds++;
+      plan.groups.back().n_upds++;
     }
   }
-  *blob_bytes = bb;
-  return GRA_OK;
+  plan.blob_bytes = bb;
+  return r->plans.emplace(key, std::move(plan)).first->second;
 }
 
 int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
@@ -963,14 +970,10 @@ int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
     g_err = "replay window out of range";
     return GRA_ERR;
   }
-  std::vector<GroupDesc> groups;
-  std::vector<HostUpd> hu;
-  uint64_t bb;
-  int rc = build_groups(r, first, n, groups, hu, &bb);
-  if (rc != GRA_OK) return rc;
+  const TickPlan &plan = plan_for(r, first, n);
   std::lock_guard<std::mutex> lk(e->mu);
-  return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n, groups,
-                         std::move(hu), bb, false);
+  return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n,
+                         plan.groups, plan.blob_bytes, false);
 }
 
 int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
@@ -986,11 +989,7 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
     g_err = "h2d window exceeds staging";
     return GRA_ERR;
   }
-  std::vector<GroupDesc> groups;
-  std::vector<HostUpd> hu;
-  uint64_t bb;
-  int rc = build_groups(r, first, n, groups, hu, &bb);
-  if (rc != GRA_OK) return rc;
+  const TickPlan &plan = plan_for(r, first, n);
   /* rebased descs (blob offsets relative to staged window) */
   std::vector<UpdDesc> ud(n);
   for (uint64_t i = 0; i < n; i++) {
@@ -999,7 +998,7 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
   }
   std::lock_guard<std::mutex> lk(e->mu);
   return e->enqueue_tick(e->d_stage_blobs, e->d_stage_descs, (uint32_t)n,
-                         groups, std::move(hu), bb, true, r->h_arena + lo,
+                         plan.groups, plan.blob_bytes, true, r->h_arena + lo,
                          hi - lo, e->d_stage_blobs, ud.data());
 }
 
