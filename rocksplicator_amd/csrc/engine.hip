@@ -27,6 +27,8 @@
  */
 #include <hip/hip_runtime.h>
 
+#define WB_UNALIGNED_OK 1 /* gfx950: misaligned global loads OK (micro_copy v1) */
+
 #include <algorithm>
 #include <atomic>
 #include <cstdio>
@@ -101,18 +103,25 @@ __host__ __device__ inline uint2 add2(uint2 a, uint2 b) {
  * memset; the fused scan2 kernel re-zeroes slot tick+32 ahead). */
 constexpr uint32_t kErrRing = 64;
 
+constexpr uint32_t kRecCache = 2; /* records cached per update by decode so
+                                     emit can skip the second blob walk */
+
 __global__ void k_decode(const uint8_t *__restrict__ blobs,
                          const UpdDesc *__restrict__ descs, uint32_t n,
                          wb::WalkTotals *__restrict__ totals, uint32_t max_rec,
                          uint32_t *__restrict__ err_ring, uint32_t tick,
-                         uint2 *__restrict__ partial,
-                         uint2 *__restrict__ bsums) {
+                         uint2 *__restrict__ partial, uint2 *__restrict__ bsums,
+                         wb::Rec *__restrict__ reccache) {
   __shared__ uint2 sh[256];
   uint32_t i = blockIdx.x * 256 + threadIdx.x;
   uint2 v = make_uint2(0, 0);
   if (i < n) {
     UpdDesc d = descs[i];
-    wb::WalkTotals t = wb::walk_f(blobs + d.off, d.len, wb::NullEmit{});
+    wb::Rec *cache = reccache + (size_t)i * kRecCache;
+    wb::WalkTotals t = wb::walk_f(blobs + d.off, d.len,
+                                  [&](const wb::Rec &r, uint32_t idx) {
+                                    if (idx < kRecCache) cache[idx] = r;
+                                  });
     if (t.n_records > max_rec) t.ok = 0;
     totals[i] = t;
     if (!t.ok) atomicAdd(&err_ring[tick % kErrRing], 1u);
@@ -191,34 +200,27 @@ __global__ void k_scan2(uint2 *__restrict__ bsums, uint32_t nblocks,
   }
 }
 
-__global__ void k_scan3(const uint2 *__restrict__ partial,
-                        const uint2 *__restrict__ bsums, uint32_t n,
-                        uint2 *__restrict__ out, uint32_t nblocks) {
-  uint32_t i = blockIdx.x * 256 + threadIdx.x;
-  if (i < n) out[i] = add2(partial[i], bsums[i / 256]);
-  if (i == n) out[n] = bsums[nblocks];
-}
-
 __global__ void k_emit(const uint8_t *__restrict__ blobs,
                        const UpdDesc *__restrict__ descs, uint32_t n,
                        const wb::WalkTotals *__restrict__ totals,
-                       const uint2 *__restrict__ scan,
+                       const uint2 *__restrict__ partial,
+                       const uint2 *__restrict__ bsums,
                        const TickPlace *__restrict__ place,
                        uint8_t *__restrict__ store,
-                       CopyTask *__restrict__ tasks) {
+                       CopyTask *__restrict__ tasks,
+                       const wb::Rec *__restrict__ reccache) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   if (place->overflow) return;
   wb::WalkTotals t = totals[i];
   if (!t.ok || t.n_records == 0) return;
   UpdDesc d = descs[i];
-  uint2 base = scan[i];
+  uint2 base = add2(partial[i], bsums[i >> 8]); /* exclusive scan, inline */
   wb::RecHdr *hdrs = (wb::RecHdr *)(store + place->hdr_off);
   uint8_t *pay_region = store + place->payload_off;
   uint32_t pay = base.y;
   uint32_t rec = base.x;
-  const uint8_t *rep = blobs + d.off;
-  wb::walk_f(rep, d.len, [&](const wb::Rec &r, uint32_t idx) {
+  auto emit_one = [&](const wb::Rec &r, uint32_t idx) {
     uint32_t cf4 = r.cf_id ? 4u : 0u;
     wb::RecHdr h;
     h.seq = d.base_seq + idx;
@@ -242,7 +244,13 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     tk.nbytes = r.val_len;
     tasks[2 * (rec + idx) + 1] = tk;
     pay += (cf4 + r.key_len + r.val_len + 15u) & ~15u;
-  });
+  };
+  if (t.n_records <= kRecCache) { /* decode cached these — no blob re-walk */
+    const wb::Rec *cache = reccache + (size_t)i * kRecCache;
+    for (uint32_t idx = 0; idx < t.n_records; idx++) emit_one(cache[idx], idx);
+  } else {
+    wb::walk_f(blobs + d.off, d.len, emit_one);
+  }
 }
 
 /* G-lane-group cooperative copy, dword funnel for misaligned sources (blob
@@ -326,14 +334,19 @@ __global__ void __launch_bounds__(256) k_copy(const uint8_t *__restrict__ blobs,
 __global__ void k_rundesc(const GroupDesc *__restrict__ groups, uint32_t ngroups,
                           const UpdDesc *__restrict__ descs,
                           const wb::WalkTotals *__restrict__ totals,
-                          const uint2 *__restrict__ scan,
+                          const uint2 *__restrict__ partial,
+                          const uint2 *__restrict__ bsums, uint32_t n,
+                          uint32_t nblocks,
                           const TickPlace *__restrict__ place,
                           DevRunDesc *__restrict__ out) {
   uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
   if (g >= ngroups) return;
   GroupDesc gd = groups[g];
-  uint2 s0 = scan[gd.first];
-  uint2 s1 = scan[gd.first + gd.n_upds];
+  auto scan_at = [&](uint32_t j) {
+    return j >= n ? bsums[nblocks] : add2(partial[j], bsums[j >> 8]);
+  };
+  uint2 s0 = scan_at(gd.first);
+  uint2 s1 = scan_at(gd.first + gd.n_upds);
   DevRunDesc rd = {};
   rd.shard = gd.shard;
   rd.n_entries = s1.x - s0.x;
@@ -399,7 +412,8 @@ struct GraEngine {
   uint32_t task_cap;
   uint32_t group_cap;        /* max contiguous shard-groups per tick */
   wb::WalkTotals *d_totals = nullptr;
-  uint2 *d_partial = nullptr, *d_bsums = nullptr, *d_scan = nullptr;
+  uint2 *d_partial = nullptr, *d_bsums = nullptr;
+  wb::Rec *d_reccache = nullptr;
   CopyTask *d_tasks = nullptr;
   TickPlace *d_place = nullptr;
   GroupDesc *d_groups = nullptr;
@@ -497,7 +511,7 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMalloc(&d_totals, (size_t)max_upd * sizeof(wb::WalkTotals)));
   HIP_TRY(hipMalloc(&d_partial, (size_t)max_upd * sizeof(uint2)));
   HIP_TRY(hipMalloc(&d_bsums, ((size_t)max_upd / 256 + 2) * sizeof(uint2)));
-  HIP_TRY(hipMalloc(&d_scan, ((size_t)max_upd + 1) * sizeof(uint2)));
+  HIP_TRY(hipMalloc(&d_reccache, (size_t)max_upd * 2 * sizeof(wb::Rec)));
   HIP_TRY(hipMalloc(&d_tasks, (size_t)task_cap * sizeof(CopyTask)));
   HIP_TRY(hipMalloc(&d_place, sizeof(TickPlace)));
   HIP_TRY(hipMalloc(&d_groups, (size_t)group_cap * sizeof(GroupDesc)));
@@ -543,7 +557,7 @@ GraEngine::~GraEngine() {
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
   for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_totals,
-                  (void *)d_partial, (void *)d_bsums, (void *)d_scan,
+                  (void *)d_partial, (void *)d_bsums, (void *)d_reccache,
                   (void *)d_tasks, (void *)d_place, (void *)d_groups,
                   (void *)d_err_ring, (void *)d_stage_blobs,
                   (void *)d_stage_descs})
@@ -620,18 +634,17 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   uint32_t nb = (n + 255) / 256;
   hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, opts.max_wb_records, d_err_ring, tick,
-                     d_partial, d_bsums);
+                     d_partial, d_bsums, d_reccache);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(t.ev[2], stream)); /* after decode(+scan1) */
   hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, d_bsums, nb,
                      d_cursor, d_place, opts.store_bytes, opts.store_ring,
                      task_cap, d_err_ring, tick);
-  hipLaunchKernelGGL(k_scan3, dim3((n + 256) / 256), dim3(256), 0, stream,
-                     d_partial, d_bsums, n, d_scan, nb);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(t.ev[3], stream)); /* after scan(+reserve) */
   hipLaunchKernelGGL(k_emit, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
-                     n, d_totals, d_scan, d_place, d_store, d_tasks);
+                     n, d_totals, d_partial, d_bsums, d_place, d_store,
+                     d_tasks, d_reccache);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(t.ev[4], stream)); /* after emit */
   /* group width by average update size (micro_copy.hip: g32 wins >=512B) */
@@ -646,7 +659,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   HIP_TRY(hipEventRecord(t.ev[5], stream)); /* after copy */
   hipLaunchKernelGGL(k_rundesc, dim3((ngroups + 255) / 256), dim3(256), 0,
                      stream, groups_for_kernel, ngroups, d_descw, d_totals,
-                     d_scan, d_place, sl.d_rundescs);
+                     d_partial, d_bsums, n, nb, d_place, sl.d_rundescs);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(t.ev[5 + 1], stream)); /* main-stream tick end */
   /* publish run descriptors on the copyout stream, overlapped with the next

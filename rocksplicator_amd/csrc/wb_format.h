@@ -42,14 +42,23 @@ enum Tag : uint8_t {
 constexpr uint32_t kHeaderBytes = 12;
 
 WB_HD uint64_t fixed64_le(const uint8_t *p) {
-  /* byte loads: rep blobs have no alignment guarantee */
+#if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
+  /* gfx950 global loads tolerate misaligned multi-dword accesses (verified
+   * by scripts/micro_copy.hip v1); blob arenas are over-allocated by 16 B */
+  return *(const uint64_t *)p;
+#else
   uint64_t v = 0;
   for (int i = 0; i < 8; i++) v |= (uint64_t)p[i] << (8 * i);
   return v;
+#endif
 }
 WB_HD uint32_t fixed32_le(const uint8_t *p) {
+#if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
+  return *(const uint32_t *)p;
+#else
   return (uint32_t)p[0] | ((uint32_t)p[1] << 8) | ((uint32_t)p[2] << 16) |
          ((uint32_t)p[3] << 24);
+#endif
 }
 
 /* varint32 decode; returns bytes consumed (1..5), 0 on error/overrun.

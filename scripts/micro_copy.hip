@@ -96,6 +96,37 @@ __device__ void copy_v2(uint8_t *dst, const uint8_t *src, uint32_t n, uint32_t l
   }
 }
 
+/* v3: 32B per lane (2x dwordx4 stores), funnel gather */
+template <int G>
+__device__ void copy_v3(uint8_t *dst, const uint8_t *src, uint32_t n, uint32_t lane) {
+  if (((uintptr_t)dst & 15) == 0) {
+    uint32_t r = (uint32_t)((uintptr_t)src & 3);
+    const uint32_t *asrc = (const uint32_t *)(src - r);
+    uint4 *d4 = (uint4 *)dst;
+    uint32_t nc2 = n >> 5; /* 32B chunks */
+    uint32_t sh = 8 * r, ish = 32 - sh;
+    for (uint32_t c2 = lane; c2 < nc2; c2 += G) {
+      uint32_t w = c2 * 8;
+      uint32_t a[9];
+#pragma unroll
+      for (int j = 0; j < 9; j++) a[j] = asrc[w + j];
+      if (r == 0) {
+        d4[2 * c2] = make_uint4(a[0], a[1], a[2], a[3]);
+        d4[2 * c2 + 1] = make_uint4(a[4], a[5], a[6], a[7]);
+      } else {
+        d4[2 * c2] = make_uint4((a[0] >> sh) | (a[1] << ish), (a[1] >> sh) | (a[2] << ish),
+                                (a[2] >> sh) | (a[3] << ish), (a[3] >> sh) | (a[4] << ish));
+        d4[2 * c2 + 1] = make_uint4((a[4] >> sh) | (a[5] << ish), (a[5] >> sh) | (a[6] << ish),
+                                    (a[6] >> sh) | (a[7] << ish), (a[7] >> sh) | (a[8] << ish));
+      }
+    }
+    uint32_t done = nc2 << 5;
+    for (uint32_t b = done + lane; b < n; b += G) dst[b] = src[b];
+  } else {
+    copy_v0<G>(dst, src, n, lane);
+  }
+}
+
 template <int V, int G>
 __global__ void __launch_bounds__(256) k_bench(const uint8_t *__restrict__ src,
                                                uint8_t *__restrict__ dst,
@@ -109,6 +140,7 @@ __global__ void __launch_bounds__(256) k_bench(const uint8_t *__restrict__ src,
     if (V == 0) copy_v0<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
     if (V == 1) copy_v1<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
     if (V == 2) copy_v2<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
+    if (V == 3) copy_v3<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
   }
 }
 
@@ -165,12 +197,13 @@ int main(int argc, char **argv) {
   };
 
   uint32_t nt = (uint32_t)h_tasks.size();
-  dim3 grid(2048), blk(256);
-#define RUN(V, G) run("v" #V "/g" #G, [&] { \
-    hipLaunchKernelGGL((k_bench<V, G>), grid, blk, 0, 0, d_src, d_dst, d_tasks, nt); })
-  RUN(0, 16); RUN(0, 32); RUN(0, 64);
-  RUN(1, 16); RUN(1, 32); RUN(1, 64);
-  RUN(2, 16); RUN(2, 32); RUN(2, 64);
-  RUN(0, 16);
+  dim3 blk(256);
+#define RUNG(V, G, NB) run("v" #V "/g" #G "/b" #NB, [&] { \
+    hipLaunchKernelGGL((k_bench<V, G>), dim3(NB), blk, 0, 0, d_src, d_dst, d_tasks, nt); })
+  RUNG(2, 16, 2048); RUNG(2, 32, 2048);
+  RUNG(3, 16, 2048); RUNG(3, 32, 2048); RUNG(3, 64, 2048);
+  RUNG(2, 32, 1024); RUNG(2, 32, 4096); RUNG(2, 32, 8192);
+  RUNG(3, 32, 1024); RUNG(3, 32, 4096); RUNG(3, 32, 8192);
+  RUNG(3, 16, 4096);
   return 0;
 }
