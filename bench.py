@@ -50,6 +50,11 @@ def parse_args():
                    help="optional rocprofv3-derived per-launch HBM traffic (see profiles/)")
     p.add_argument("--h2d", action="store_true",
                    help="measure the PCIe-inclusive staging path instead (side report)")
+    p.add_argument("--repartition", action="store_true",
+                   help="BASELINE config #4: re-home update blobs to their "
+                        "owner rank via RCCL all-to-all over xGMI inside "
+                        "each timed step (at world=1: device self-copy, "
+                        "same code path)")
     return p.parse_args()
 
 
@@ -101,14 +106,111 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s):
     }
 
 
+def run_repartition(args, rank, world, local, dist):
+    """Config #4 path: per timed step, an all-to-all re-homes one tick of
+    update blobs to their owner ranks over RCCL/xGMI, then the owner applies
+    them from the received device buffer (zero-copy)."""
+    import torch
+
+    from rocksplicator_amd import repartition as rp
+
+    torch.cuda.set_device(local)
+    seed_base = 0xB0CC5EED
+    send_bytes, in_splits = rp.build_send(
+        rank, world, args.nshards, args.tick_updates, args.key_len,
+        args.val_len, args.kind, seed_base)
+    out_splits, cdescs, n_recv, _ = rp.expected_recv(
+        rank, world, args.nshards, args.tick_updates, args.key_len,
+        args.val_len, args.kind, seed_base)
+    send_t = torch.frombuffer(bytearray(send_bytes), dtype=torch.uint8).cuda()
+    recv_t = torch.empty(sum(out_splits) + 16, dtype=torch.uint8, device="cuda")
+
+    eng = ra.Engine(nshards=args.nshards, device=-1, store_ring=1,
+                    store_bytes=min(int(sum(out_splits) * 2.6) + (1 << 30), 48 << 30))
+    rep = rp.upload_dev(eng, recv_t.data_ptr(), sum(out_splits), cdescs, n_recv)
+
+    def exchange():
+        if dist is not None:
+            import torch.distributed as td
+            td.all_to_all_single(recv_t[:sum(out_splits)], send_t,
+                                 out_splits, in_splits)
+        else:  # world 1: self-exchange, same downstream path
+            recv_t[:len(send_bytes)].copy_(send_t)
+        torch.cuda.synchronize()
+
+    def one_step(_i):
+        exchange()
+        rep.tick(0, n_recv)
+
+    for i in range(args.warmup):
+        one_step(i)
+    rep.sync()
+    eng.stats_reset()
+    if dist:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    rep.sync()
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    if dist:
+        dist.barrier()
+    wall = t1 - t0
+    if dist:
+        tw = torch.tensor([wall], dtype=torch.float64)
+        dist.all_reduce(tw, op=dist.ReduceOp.MAX)
+        wall = float(tw.item())
+    return eng, wall, n_recv
+
+
 def main():
     args = parse_args()
     rank, world, local = dist_env()
     dist = None
     if world > 1:
         import torch.distributed as td
-        td.init_process_group(backend="gloo")
+        # gloo for control-plane barriers; nccl (RCCL) group for the
+        # repartition all-to-all when enabled
+        td.init_process_group(backend="nccl" if args.repartition else "gloo")
         dist = td
+
+    if args.repartition:
+        eng, wall, n_recv = run_repartition(args, rank, world, local, dist)
+        st = eng.stats()
+        value = args.steps * n_recv * world / wall
+        result = {
+            "metric": "replicated updates/sec (Put+Delete) at 1024 shards",
+            "value": value,
+            "unit": "updates/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": wall * 1e3 / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": f"{args.nshards}shards/gpu_{args.key_len}Bkey_"
+                            f"{args.val_len}Bval_repartition",
+                "nshards_per_gpu": args.nshards,
+                "tick_updates": args.tick_updates,
+                "path": "rccl-alltoall-repartition + hbm apply",
+                "parallelism": f"shard-parallel x{world} + all-to-all",
+            },
+            "kernels_ms_per_tick": {
+                "copy": st.copy_ms / max(st.ticks, 1),
+                "total": st.total_ms / max(st.ticks, 1),
+            },
+        }
+        if rank == 0:
+            print(json.dumps(result))
+        eng.close()
+        if dist:
+            dist.destroy_process_group()
+        return
 
     # --- generate the replay stream (untimed) ---
     n_ticks_data = min(args.warmup + args.steps, args.max_ticks_resident)

@@ -145,6 +145,13 @@ void gra_pin_free(GraEngine *e, uint8_t *ptr);
  * shard's current seq. */
 int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
                const GraUpdateDesc *descs, uint64_t ndescs, GraReplay **out);
+/* Same, but the blobs already live in device memory (e.g. the output tensor
+ * of an RCCL all-to-all shard repartition — BASELINE config #4): zero-copy,
+ * caller owns the arena (>=16 B readable slack required) and supplies each
+ * batch's header record count (host can't read device headers). */
+int gra_upload_dev(GraEngine *e, void *dev_arena, size_t arena_bytes,
+                   const GraUpdateDesc *descs, uint64_t ndescs,
+                   const uint32_t *counts, GraReplay **out);
 void gra_replay_destroy(GraReplay *r);
 
 /* Run one pipeline tick over descs [first, first+n): decode + emit +

@@ -467,6 +467,7 @@ struct TickPlan {
 struct GraReplay {
   GraEngine *e = nullptr;
   uint8_t *d_blobs = nullptr;
+  bool external_blobs = false; /* caller-owned device arena (gra_upload_dev) */
   UpdDesc *d_descs = nullptr;
   std::vector<UpdDesc> descs; /* host copy */
   std::vector<uint16_t> counts;
@@ -959,23 +960,14 @@ void gra_pin_free(GraEngine *e, uint8_t *ptr) {
   (void)hipHostFree(ptr);
 }
 
-int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
-               const GraUpdateDesc *descs, uint64_t n, GraReplay **out) {
-  auto *r = new GraReplay();
-  r->e = e;
-  r->arena_bytes = arena_bytes;
-  r->h_arena = arena;
-  hipError_t rc1 = hipMalloc(&r->d_blobs, arena_bytes + 16);
-  hipError_t rc2 = hipMalloc(&r->d_descs, (size_t)n * sizeof(UpdDesc));
-  if (rc1 != hipSuccess || rc2 != hipSuccess) {
-    g_err = "gra_upload: device allocation failed";
-    delete r;
-    return GRA_ERR;
-  }
-  if (hipMemcpy(r->d_blobs, arena, arena_bytes, hipMemcpyHostToDevice) !=
-      hipSuccess) {
-    g_err = "gra_upload: H2D failed";
-    delete r;
+/* host_counts: batch record counts read from a HOST copy of the headers
+ * (needed because with an external device arena the host cannot read the
+ * blob bytes); pass nullptr when `arena` is host-readable. */
+static int upload_common(GraEngine *e, GraReplay *r, const uint8_t *arena,
+                         size_t arena_bytes, const GraUpdateDesc *descs,
+                         uint64_t n, const uint32_t *host_counts) {
+  if (hipMalloc(&r->d_descs, (size_t)n * sizeof(UpdDesc)) != hipSuccess) {
+    g_err = "gra_upload: desc allocation failed";
     return GRA_ERR;
   }
   r->descs.resize(n);
@@ -985,10 +977,10 @@ int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
     if (d.shard >= e->opts.nshards || d.off + d.len > arena_bytes ||
         d.len < wb::kHeaderBytes) {
       g_err = "gra_upload: bad desc";
-      delete r;
       return GRA_ERR;
     }
-    uint32_t count = wb::fixed32_le(arena + d.off + 8);
+    uint32_t count = host_counts ? host_counts[i]
+                                 : wb::fixed32_le(arena + d.off + 8);
     ShardState &ss = e->shards[d.shard];
     UpdDesc u;
     u.off = d.off;
@@ -1002,8 +994,49 @@ int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
   if (hipMemcpy(r->d_descs, r->descs.data(), (size_t)n * sizeof(UpdDesc),
                 hipMemcpyHostToDevice) != hipSuccess) {
     g_err = "gra_upload: desc H2D failed";
+    return GRA_ERR;
+  }
+  return GRA_OK;
+}
+
+int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
+               const GraUpdateDesc *descs, uint64_t n, GraReplay **out) {
+  auto *r = new GraReplay();
+  r->e = e;
+  r->arena_bytes = arena_bytes;
+  r->h_arena = arena;
+  if (hipMalloc(&r->d_blobs, arena_bytes + 16) != hipSuccess ||
+      hipMemcpy(r->d_blobs, arena, arena_bytes, hipMemcpyHostToDevice) !=
+          hipSuccess) {
+    g_err = "gra_upload: blob allocation/H2D failed";
     delete r;
     return GRA_ERR;
+  }
+  int rc = upload_common(e, r, arena, arena_bytes, descs, n, nullptr);
+  if (rc != GRA_OK) {
+    delete r;
+    return rc;
+  }
+  *out = r;
+  return GRA_OK;
+}
+
+/* Device-resident arena (e.g. an RCCL all-to-all output tensor): the engine
+ * reads blobs in place, zero-copy. The caller owns dev_arena (must outlive
+ * the replay and include >=16 B of readable slack past arena_bytes) and
+ * passes batch record counts explicitly. */
+int gra_upload_dev(GraEngine *e, void *dev_arena, size_t arena_bytes,
+                   const GraUpdateDesc *descs, uint64_t n,
+                   const uint32_t *counts, GraReplay **out) {
+  auto *r = new GraReplay();
+  r->e = e;
+  r->arena_bytes = arena_bytes;
+  r->d_blobs = (uint8_t *)dev_arena;
+  r->external_blobs = true;
+  int rc = upload_common(e, r, nullptr, arena_bytes, descs, n, counts);
+  if (rc != GRA_OK) {
+    delete r;
+    return rc;
   }
   *out = r;
   return GRA_OK;
@@ -1011,7 +1044,7 @@ int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
 
 void gra_replay_destroy(GraReplay *r) {
   if (!r) return;
-  if (r->d_blobs) (void)hipFree(r->d_blobs);
+  if (r->d_blobs && !r->external_blobs) (void)hipFree(r->d_blobs);
   if (r->d_descs) (void)hipFree(r->d_descs);
   for (auto &kv : r->plans)
     if (kv.second.d_groups) (void)hipFree(kv.second.d_groups);
