@@ -167,6 +167,14 @@ def run_repartition(args, rank, world, local, dist):
 def main():
     args = parse_args()
     rank, world, local = dist_env()
+    if args.gpus > 1 and world == 1:
+        # invoked without torchrun: self-launch one rank per GPU
+        import subprocess
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
+               "--master-port", "29711", os.path.abspath(__file__),
+               *[a for a in sys.argv[1:]]]
+        raise SystemExit(subprocess.call(cmd))
     dist = None
     if world > 1:
         import torch.distributed as td

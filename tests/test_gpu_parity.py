@@ -235,3 +235,34 @@ def test_mixed_leader_and_replicated(olib):
     for k in (b"a", b"b"):
         assert db.get(k) == ost.get(0, k)
     e.close()
+
+
+def test_full_config_shape_parity(olib):
+    """Config #3 at a size the oracle still finishes in seconds (1024 shards,
+    16B/1KB Zipf): per-shard seq equality for ALL shards + sampled per-key
+    Get equality + the size-independent invariant that total applied records
+    equals total seq advance (a6)."""
+    nshards, nupd = 1024, 102400
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=1024, kind=1,
+                                       seed=2024)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd)
+    e = ra.Engine(nshards=nshards, store_bytes=4 << 30)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    for first in range(0, nupd, 51200):
+        rep.tick(first, 51200)
+    rep.sync()
+    st = e.stats()
+    assert st.records == nupd
+    total_seq = 0
+    for s in range(nshards):
+        db = e.open(s)
+        seq = db.latest_seq()
+        assert seq == ost.latest_seq(s), f"shard {s}"
+        total_seq += seq
+        db.close()
+    assert total_seq == nupd  # every update consumed exactly one seq
+    keys = collect_keys(olib, raw, descs, 6000, per_shard_cap=8)
+    check_parity(e, ost, keys, range(0, nshards, 61))
+    e.close()
