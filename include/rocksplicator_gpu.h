@@ -152,6 +152,21 @@ int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
 int gra_upload_dev(GraEngine *e, void *dev_arena, size_t arena_bytes,
                    const GraUpdateDesc *descs, uint64_t ndescs,
                    const uint32_t *counts, GraReplay **out);
+/* Config #5: Snappy-compressed Update payloads. descs index the COMPRESSED
+ * arena; ulens/counts carry per-update uncompressed size and batch record
+ * count. A GPU pre-stage (k_snappy, lane-per-update) decompresses each tick
+ * into a scratch arena feeding the normal decode pipeline. */
+int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
+                      size_t comp_bytes, const GraUpdateDesc *descs,
+                      uint64_t ndescs, const uint32_t *ulens,
+                      const uint32_t *counts, GraReplay **out);
+/* Host-side Snappy codec (transport/leader side + harness). Returns
+ * compressed length (0 = dst too small) / uncompressed length (UINT32_MAX =
+ * corrupt). */
+uint32_t gra_snappy_compress(const uint8_t *src, uint32_t slen, uint8_t *dst,
+                             uint32_t dcap);
+uint32_t gra_snappy_decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
+                               uint32_t dcap);
 void gra_replay_destroy(GraReplay *r);
 
 /* Run one pipeline tick over descs [first, first+n): decode + emit +
@@ -167,7 +182,8 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n);
 
 /* ---------------- stats / timing (HIP-event-measured, per kernel) -------- */
 typedef struct {
-  double h2d_ms, decode_ms, scan_ms, emit_ms, copy_ms, runfix_ms, total_ms;
+  double h2d_ms, snappy_ms, decode_ms, scan_ms, emit_ms, copy_ms, runfix_ms,
+      total_ms;
   uint64_t ticks, updates, records, blob_bytes, payload_bytes;
 } GraStats;
 void gra_stats(GraEngine *e, GraStats *out);
@@ -189,6 +205,8 @@ typedef struct {
   uint64_t key_space;      /* e.g. 1<<24 */
   double zipf_s;           /* e.g. 0.99 */
   uint64_t seed;
+  uint32_t compressible;   /* 1: low-entropy values (Snappy-compressible) */
+  uint32_t _pad;
 } GraGenOpts;
 int gra_gen_stream(const GraGenOpts *g, uint64_t n_updates, uint8_t *arena,
                    size_t arena_cap, size_t *arena_used, GraUpdateDesc *descs,

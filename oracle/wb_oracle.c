@@ -561,3 +561,171 @@ double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena, const OrcUpdateDes
   clock_gettime(CLOCK_MONOTONIC, &t1);
   return (t1.tv_sec - t0.tv_sec) + (t1.tv_nsec - t0.tv_nsec) * 1e-9;
 }
+
+/* ================= Snappy block format ================= */
+size_t orc_snappy_max_len(size_t n) { return 32 + n + n / 6; }
+
+static size_t snp_varint(uint8_t *dst, uint32_t v) {
+  size_t i = 0;
+  while (v >= 0x80) {
+    dst[i++] = (uint8_t)(v | 0x80);
+    v >>= 7;
+  }
+  dst[i++] = (uint8_t)v;
+  return i;
+}
+
+static size_t snp_emit_literal(uint8_t *dst, const uint8_t *src, size_t len) {
+  size_t i = 0;
+  size_t n = len - 1;
+  if (n < 60) {
+    dst[i++] = (uint8_t)(n << 2);
+  } else if (n < (1u << 8)) {
+    dst[i++] = 60 << 2;
+    dst[i++] = (uint8_t)n;
+  } else if (n < (1u << 16)) {
+    dst[i++] = 61 << 2;
+    dst[i++] = (uint8_t)n;
+    dst[i++] = (uint8_t)(n >> 8);
+  } else if (n < (1u << 24)) {
+    dst[i++] = 62 << 2;
+    dst[i++] = (uint8_t)n;
+    dst[i++] = (uint8_t)(n >> 8);
+    dst[i++] = (uint8_t)(n >> 16);
+  } else {
+    dst[i++] = 63 << 2;
+    dst[i++] = (uint8_t)n;
+    dst[i++] = (uint8_t)(n >> 8);
+    dst[i++] = (uint8_t)(n >> 16);
+    dst[i++] = (uint8_t)(n >> 24);
+  }
+  memcpy(dst + i, src, len);
+  return i + len;
+}
+
+size_t orc_snappy_compress(const uint8_t *src, size_t slen, uint8_t *dst,
+                           size_t dcap) {
+  if (dcap < orc_snappy_max_len(slen)) return 0;
+  size_t o = snp_varint(dst, (uint32_t)slen);
+  enum { HBITS = 13, HSIZE = 1 << HBITS };
+  uint32_t tab[HSIZE];
+  memset(tab, 0xFF, sizeof(tab));
+  size_t pos = 0, lit = 0;
+  while (pos + 4 <= slen) {
+    uint32_t cur;
+    memcpy(&cur, src + pos, 4);
+    uint32_t h = (cur * 0x1e35a7bdu) >> (32 - HBITS);
+    uint32_t cand = tab[h];
+    tab[h] = (uint32_t)pos;
+    uint32_t c4;
+    if (cand != 0xFFFFFFFFu && pos - cand <= 0xFFFF &&
+        (memcpy(&c4, src + cand, 4), c4 == cur)) {
+      if (pos > lit) o += snp_emit_literal(dst + o, src + lit, pos - lit);
+      size_t len = 4, maxlen = slen - pos;
+      if (maxlen > 64) maxlen = 64;
+      while (len < maxlen && src[cand + len] == src[pos + len]) len++;
+      uint32_t off = (uint32_t)(pos - cand);
+      dst[o++] = (uint8_t)(((len - 1) << 2) | 2); /* 2-byte-offset copy */
+      dst[o++] = (uint8_t)off;
+      dst[o++] = (uint8_t)(off >> 8);
+      pos += len;
+      lit = pos;
+    } else {
+      pos++;
+    }
+  }
+  if (slen > lit) o += snp_emit_literal(dst + o, src + lit, slen - lit);
+  return o;
+}
+
+int orc_snappy_decompress(const uint8_t *src, size_t slen, uint8_t *dst,
+                          size_t dcap, size_t *dlen) {
+  uint32_t ulen;
+  size_t ip = varint32_decode(src, src + slen, &ulen);
+  if (ip == 0 || ulen > dcap) return 1;
+  size_t op = 0;
+  while (ip < slen) {
+    uint8_t tag = src[ip++];
+    if ((tag & 3) == 0) { /* literal */
+      uint32_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        uint32_t nb = len - 60;
+        if (ip + nb > slen) return 2;
+        len = 0;
+        for (uint32_t b = 0; b < nb; b++) len |= (uint32_t)src[ip + b] << (8 * b);
+        len += 1;
+        ip += nb;
+      }
+      if (ip + len > slen || op + len > ulen) return 3;
+      memcpy(dst + op, src + ip, len);
+      ip += len;
+      op += len;
+    } else {
+      uint32_t len, off;
+      if ((tag & 3) == 1) {
+        len = ((tag >> 2) & 7) + 4;
+        if (ip + 1 > slen) return 4;
+        off = ((uint32_t)(tag >> 5) << 8) | src[ip];
+        ip += 1;
+      } else if ((tag & 3) == 2) {
+        len = (tag >> 2) + 1;
+        if (ip + 2 > slen) return 4;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+      } else {
+        len = (tag >> 2) + 1;
+        if (ip + 4 > slen) return 4;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+              ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > op || op + len > ulen) return 5;
+      for (uint32_t b = 0; b < len; b++) dst[op + b] = dst[op + b - off];
+      op += len;
+    }
+  }
+  if (op != ulen) return 6;
+  *dlen = op;
+  return 0;
+}
+
+typedef struct {
+  OrcStore *s;
+  const uint8_t *arena;
+  const OrcUpdateDesc *descs;
+  uint64_t n;
+  int tid, nthreads;
+} SnapBenchArg;
+
+static void *snap_bench_worker(void *p) {
+  SnapBenchArg *a = (SnapBenchArg *)p;
+  uint8_t *scratch = (uint8_t *)malloc(1 << 20);
+  for (uint64_t i = 0; i < a->n; i++) {
+    const OrcUpdateDesc *d = &a->descs[i];
+    if ((int)(d->shard % (uint32_t)a->nthreads) != a->tid) continue;
+    size_t ulen;
+    if (orc_snappy_decompress(a->arena + d->off, d->len, scratch, 1 << 20,
+                              &ulen) == 0)
+      orc_apply(a->s, d->shard, scratch, ulen, d->ts);
+  }
+  free(scratch);
+  return NULL;
+}
+
+double orc_cpu_snappy_apply_bench(OrcStore *s, const uint8_t *arena,
+                                  const OrcUpdateDesc *descs, uint64_t n,
+                                  int nthreads) {
+  if (nthreads < 1) nthreads = 1;
+  if (nthreads > 256) nthreads = 256;
+  pthread_t th[256];
+  SnapBenchArg args[256];
+  struct timespec t0, t1;
+  clock_gettime(CLOCK_MONOTONIC, &t0);
+  for (int i = 0; i < nthreads; i++) {
+    args[i] = (SnapBenchArg){s, arena, descs, n, i, nthreads};
+    pthread_create(&th[i], NULL, snap_bench_worker, &args[i]);
+  }
+  for (int i = 0; i < nthreads; i++) pthread_join(th[i], NULL);
+  clock_gettime(CLOCK_MONOTONIC, &t1);
+  return (t1.tv_sec - t0.tv_sec) + (t1.tv_nsec - t0.tv_nsec) * 1e-9;
+}

@@ -121,7 +121,23 @@ typedef struct {
 double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena,
                            const OrcUpdateDesc *descs, uint64_t n, int nthreads);
 
+
+/* ---------- Snappy block format (transport compression, config #5) ----------
+ * Independent CPU restatement of the public Snappy format (raw block
+ * format: varint32 uncompressed length, then literal/copy elements).
+ * Checker-side codec for the GPU decompress stage. */
+size_t orc_snappy_max_len(size_t n);
+/* returns compressed size, 0 on overflow/cap */
+size_t orc_snappy_compress(const uint8_t *src, size_t slen, uint8_t *dst, size_t dcap);
+/* returns 0 ok (sets *dlen), nonzero on corruption */
+int orc_snappy_decompress(const uint8_t *src, size_t slen, uint8_t *dst,
+                          size_t dcap, size_t *dlen);
+/* CPU baseline variant: decompress each update then apply (config #5). */
+double orc_cpu_snappy_apply_bench(OrcStore *s, const uint8_t *arena,
+                                  const OrcUpdateDesc *descs, uint64_t n,
+                                  int nthreads);
+
 #ifdef __cplusplus
 }
 #endif
-#endif
+#endif /* WB_ORACLE_H */

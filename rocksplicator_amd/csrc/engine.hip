@@ -42,6 +42,7 @@
 
 #include "../../include/rocksplicator_gpu.h"
 #include "host_store.h"
+#include "snappy.h"
 #include "wb_format.h"
 
 /* ---------------- error plumbing ---------------- */
@@ -79,6 +80,11 @@ struct CopyTask {
   uint32_t dst_rel; /* into tick payload region */
   uint32_t nbytes;
 };
+struct SnapTask { /* one compressed Update payload (config #5) */
+  uint64_t comp_off;
+  uint32_t comp_len;
+  uint32_t ulen;
+};
 struct GroupDesc {
   uint32_t shard, first, n_upds, _pad;
 };
@@ -96,6 +102,27 @@ __host__ __device__ inline uint2 add2(uint2 a, uint2 b) {
 }
 
 /* ---------------- kernels ---------------- */
+
+/* K0 (config #5): per-update Snappy decompress, compressed arena ->
+ * uncompressed scratch (= the tick's blob arena). Lane-per-update: each
+ * stream is sequentially dependent; parallelism comes from the 200k+
+ * updates in flight. A failed stream poisons its slot header so the decode
+ * walk rejects it. */
+__global__ void k_snappy(const uint8_t *__restrict__ comp,
+                         const SnapTask *__restrict__ tasks,
+                         const UpdDesc *__restrict__ descs, uint32_t n,
+                         uint8_t *__restrict__ scratch,
+                         uint32_t *__restrict__ err_ring, uint32_t tick) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  SnapTask t = tasks[i];
+  uint8_t *dst = scratch + descs[i].off;
+  uint32_t r = snp::decompress(comp + t.comp_off, t.comp_len, dst, t.ulen);
+  if (r != t.ulen) {
+    for (int b = 0; b < 13; b++) dst[b] = 0xFF; /* force decode rejection */
+    atomicAdd(&err_ring[tick % 64u], 1u);
+  }
+}
 
 /* K1: decode walk fused with the first scan pass — each block decodes 256
  * updates, then block-scans (records, payload16) into exclusive partials +
@@ -368,13 +395,13 @@ __global__ void k_rundesc(const GroupDesc *__restrict__ groups, uint32_t ngroups
 /* ---------------- host engine ---------------- */
 
 struct Stats {
-  double h2d_ms = 0, decode_ms = 0, scan_ms = 0, emit_ms = 0, copy_ms = 0,
-         runfix_ms = 0, total_ms = 0;
+  double h2d_ms = 0, snappy_ms = 0, decode_ms = 0, scan_ms = 0, emit_ms = 0,
+         copy_ms = 0, runfix_ms = 0, total_ms = 0;
   uint64_t ticks = 0, updates = 0, records = 0, blob_bytes = 0, payload_bytes = 0;
 };
 
 constexpr int kSlots = 8;
-constexpr int kEventsPerTick = 8;
+constexpr int kEventsPerTick = 10;
 
 struct TickRec {
   int slot = -1;
@@ -443,7 +470,9 @@ struct GraEngine {
                    bool time_h2d, const void *h2d_src = nullptr,
                    size_t h2d_bytes = 0, uint8_t *d_h2d_dst = nullptr,
                    const UpdDesc *h_descs_src = nullptr,
-                   const GroupDesc *d_groups_dev = nullptr);
+                   const GroupDesc *d_groups_dev = nullptr,
+                   const uint8_t *d_comp = nullptr,
+                   const SnapTask *d_snaptasks = nullptr);
   int ingest(bool wait_all);
   int ingest_one(TickRec &t, bool wait);
   int flush_locked();
@@ -473,6 +502,11 @@ struct GraReplay {
   std::vector<uint16_t> counts;
   size_t arena_bytes = 0;
   const uint8_t *h_arena = nullptr; /* for tick_h2d (must stay alive) */
+  /* config #5: compressed transport — d_blobs becomes the uncompressed
+   * scratch arena, blobs decompressed per tick by k_snappy */
+  uint8_t *d_comp = nullptr;
+  SnapTask *d_snaptasks = nullptr;
+  bool snappy = false;
   std::map<std::pair<uint64_t, uint64_t>, TickPlan> plans; /* window cache */
 };
 
@@ -586,7 +620,9 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                             uint64_t blob_bytes, bool time_h2d,
                             const void *h2d_src, size_t h2d_bytes,
                             uint8_t *d_h2d_dst, const UpdDesc *h_descs_src,
-                            const GroupDesc *d_groups_dev) {
+                            const GroupDesc *d_groups_dev,
+                            const uint8_t *d_comp,
+                            const SnapTask *d_snaptasks) {
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
@@ -633,6 +669,13 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   }
   HIP_TRY(hipEventRecord(t.ev[1], stream)); /* after h2d */
   uint32_t nb = (n + 255) / 256;
+  if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena */
+    hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, stream, d_comp,
+                       d_snaptasks, d_descw, n, (uint8_t *)d_blobs, d_err_ring,
+                       tick);
+    HIP_TRY(hipGetLastError());
+  }
+  HIP_TRY(hipEventRecord(t.ev[8], stream)); /* after snappy */
   hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, opts.max_wb_records, d_err_ring, tick,
                      d_partial, d_bsums, d_reccache);
@@ -696,7 +739,8 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     return (double)ms;
   };
   if (t.h2d_timed) stats.h2d_ms += dt(0, 1);
-  stats.decode_ms += dt(1, 2);
+  stats.snappy_ms += dt(1, 8);
+  stats.decode_ms += dt(8, 2);
   stats.scan_ms += dt(2, 3);
   stats.emit_ms += dt(3, 4);
   stats.copy_ms += dt(4, 5);
@@ -1042,9 +1086,62 @@ int gra_upload_dev(GraEngine *e, void *dev_arena, size_t arena_bytes,
   return GRA_OK;
 }
 
+/* Config #5 upload: Update payloads are Snappy-compressed in transit.
+ * comp descs give (off,len) into the compressed arena; ulen[i] is each
+ * update's uncompressed size (transport metadata); counts[i] the batch
+ * record count (headers unreadable on host while compressed). The engine
+ * decompresses per tick on-GPU (k_snappy) into a scratch arena that then
+ * feeds the normal decode pipeline. */
+int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
+                      size_t comp_bytes, const GraUpdateDesc *descs,
+                      uint64_t n, const uint32_t *ulens,
+                      const uint32_t *counts, GraReplay **out) {
+  auto *r = new GraReplay();
+  r->e = e;
+  r->snappy = true;
+  /* scratch layout: per-update slot, 16-B aligned, +8 B chunk-write slack */
+  std::vector<UpdDesc> tmp; /* not used; offsets computed inline */
+  std::vector<GraUpdateDesc> udescs(n);
+  std::vector<SnapTask> tasks(n);
+  uint64_t scratch = 0;
+  for (uint64_t i = 0; i < n; i++) {
+    tasks[i] = {descs[i].off, descs[i].len, ulens[i]};
+    udescs[i].shard = descs[i].shard;
+    udescs[i].len = ulens[i];
+    udescs[i].off = scratch;
+    udescs[i].ts = descs[i].ts;
+    scratch += ((uint64_t)ulens[i] + 8 + 15) & ~15ULL;
+  }
+  r->arena_bytes = scratch;
+  if (hipMalloc(&r->d_blobs, scratch + 16) != hipSuccess ||
+      hipMalloc(&r->d_comp, comp_bytes + 16) != hipSuccess ||
+      hipMalloc(&r->d_snaptasks, n * sizeof(SnapTask)) != hipSuccess) {
+    g_err = "gra_upload_snappy: allocation failed";
+    delete r;
+    return GRA_ERR;
+  }
+  if (hipMemcpy(r->d_comp, comp_arena, comp_bytes, hipMemcpyHostToDevice) !=
+          hipSuccess ||
+      hipMemcpy(r->d_snaptasks, tasks.data(), n * sizeof(SnapTask),
+                hipMemcpyHostToDevice) != hipSuccess) {
+    g_err = "gra_upload_snappy: H2D failed";
+    delete r;
+    return GRA_ERR;
+  }
+  int rc = upload_common(e, r, nullptr, scratch, udescs.data(), n, counts);
+  if (rc != GRA_OK) {
+    delete r;
+    return rc;
+  }
+  *out = r;
+  return GRA_OK;
+}
+
 void gra_replay_destroy(GraReplay *r) {
   if (!r) return;
   if (r->d_blobs && !r->external_blobs) (void)hipFree(r->d_blobs);
+  if (r->d_comp) (void)hipFree(r->d_comp);
+  if (r->d_snaptasks) (void)hipFree(r->d_snaptasks);
   if (r->d_descs) (void)hipFree(r->d_descs);
   for (auto &kv : r->plans)
     if (kv.second.d_groups) (void)hipFree(kv.second.d_groups);
@@ -1094,7 +1191,8 @@ int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
   std::lock_guard<std::mutex> lk(e->mu);
   return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n,
                          plan.groups, plan.blob_bytes, false, nullptr, 0,
-                         nullptr, nullptr, plan.d_groups);
+                         nullptr, nullptr, plan.d_groups, r->d_comp,
+                         r->snappy ? r->d_snaptasks + first : nullptr);
 }
 
 int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
@@ -1130,6 +1228,7 @@ void gra_stats(GraEngine *e, GraStats *out) {
   (void)e->ingest(false);
   GraStats s;
   s.h2d_ms = e->stats.h2d_ms;
+  s.snappy_ms = e->stats.snappy_ms;
   s.decode_ms = e->stats.decode_ms;
   s.scan_ms = e->stats.scan_ms;
   s.emit_ms = e->stats.emit_ms;

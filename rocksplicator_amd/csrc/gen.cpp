@@ -10,6 +10,7 @@
 #include <vector>
 
 #include "../../include/rocksplicator_gpu.h"
+#include "snappy.h"
 #include "wb_format.h"
 
 namespace {
@@ -135,15 +136,25 @@ extern "C" int gra_gen_stream(const GraGenOpts *g, uint64_t n_updates,
       pos += kl;
       if (tag != wb::kDeletion) {
         pos += put_varint(p + pos, this_vl);
-        /* value bytes from xoshiro, 8 at a time */
-        uint32_t i = 0;
-        for (; i + 8 <= this_vl; i += 8) {
-          uint64_t x = rng.next();
-          memcpy(p + pos + i, &x, 8);
-        }
-        if (i < this_vl) {
-          uint64_t x = rng.next();
-          memcpy(p + pos + i, &x, this_vl - i);
+        if (g->compressible) {
+          /* low-entropy values (config #5 transport-compression shape):
+           * 64-byte blocks drawn from 4 seed-derived patterns */
+          for (uint32_t i = 0; i < this_vl; i += 64) {
+            uint32_t c = (uint32_t)(rng.next() & 3);
+            uint32_t run = this_vl - i < 64 ? this_vl - i : 64;
+            for (uint32_t j = 0; j < run; j++)
+              p[pos + i + j] = (uint8_t)(c * 67 + (j & 15) * 13 + 7);
+          }
+        } else {
+          uint32_t i = 0;
+          for (; i + 8 <= this_vl; i += 8) {
+            uint64_t x = rng.next();
+            memcpy(p + pos + i, &x, 8);
+          }
+          if (i < this_vl) {
+            uint64_t x = rng.next();
+            memcpy(p + pos + i, &x, this_vl - i);
+          }
         }
         pos += this_vl;
       }
@@ -157,4 +168,13 @@ extern "C" int gra_gen_stream(const GraGenOpts *g, uint64_t n_updates,
   }
   if (arena_used) *arena_used = off;
   return GRA_OK;
+}
+
+extern "C" uint32_t gra_snappy_compress(const uint8_t *src, uint32_t slen,
+                                        uint8_t *dst, uint32_t dcap) {
+  return snp::compress(src, slen, dst, dcap);
+}
+extern "C" uint32_t gra_snappy_decompress(const uint8_t *src, uint32_t slen,
+                                          uint8_t *dst, uint32_t dcap) {
+  return snp::decompress(src, slen, dst, dcap);
 }

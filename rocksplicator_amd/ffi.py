@@ -47,6 +47,7 @@ class GraUpdateDesc(C.Structure):
 class GraStats(C.Structure):
     _fields_ = [
         ("h2d_ms", C.c_double),
+        ("snappy_ms", C.c_double),
         ("decode_ms", C.c_double),
         ("scan_ms", C.c_double),
         ("emit_ms", C.c_double),
@@ -70,6 +71,8 @@ class GraGenOpts(C.Structure):
         ("key_space", C.c_uint64),
         ("zipf_s", C.c_double),
         ("seed", C.c_uint64),
+        ("compressible", C.c_uint32),
+        ("_pad", C.c_uint32),
     ]
 
 
@@ -122,6 +125,11 @@ def load():
     lib.gra_wb_data.argtypes = [C.c_void_p, C.POINTER(C.c_size_t)]
     lib.gra_wb_data.restype = C.POINTER(C.c_uint8)
     lib.gra_gen_stream.argtypes = [C.POINTER(GraGenOpts), C.c_uint64, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(C.c_size_t), C.POINTER(GraUpdateDesc), C.c_int64]
+    lib.gra_upload_snappy.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_uint32), C.POINTER(C.c_uint32), C.POINTER(C.c_void_p)]
+    lib.gra_snappy_compress.restype = C.c_uint32
+    lib.gra_snappy_compress.argtypes = [C.c_char_p, C.c_uint32, C.c_char_p, C.c_uint32]
+    lib.gra_snappy_decompress.restype = C.c_uint32
+    lib.gra_snappy_decompress.argtypes = [C.c_char_p, C.c_uint32, C.c_char_p, C.c_uint32]
     _lib = lib
     return lib
 
@@ -237,6 +245,16 @@ class Engine:
             raise RuntimeError(f"gra_upload rc={rc}: {last_error(self.lib)}")
         return Replay(self, out)
 
+    def upload_snappy(self, comp_arena, comp_bytes, descs, n, ulens, counts=None):
+        out = C.c_void_p()
+        ul = (C.c_uint32 * n)(*ulens)
+        cn = (C.c_uint32 * n)(*(counts if counts is not None else [1] * n))
+        rc = self.lib.gra_upload_snappy(self.h, comp_arena, comp_bytes, descs,
+                                        n, ul, cn, C.byref(out))
+        if rc != GRA_OK:
+            raise RuntimeError(f"gra_upload_snappy rc={rc}: {last_error(self.lib)}")
+        return Replay(self, out)
+
     def pin_alloc(self, nbytes):
         p = C.POINTER(C.c_uint8)()
         rc = self.lib.gra_pin_alloc(self.h, nbytes, C.byref(p))
@@ -311,11 +329,12 @@ class Replay:
 
 def gen_stream(nshards, n_updates, key_len=16, val_len=128, kind=0,
                key_space=1 << 24, zipf_s=0.99, seed=0, ts=0, arena=None,
-               arena_cap=None):
+               arena_cap=None, compressible=0):
     """Generate a deterministic synthetic replay stream into `arena`
     (a ctypes pointer/buffer) — returns (arena, used_bytes, descs)."""
     lib = load()
-    g = GraGenOpts(nshards, key_len, val_len, kind, key_space, zipf_s, seed)
+    g = GraGenOpts(nshards, key_len, val_len, kind, key_space, zipf_s, seed,
+                   compressible, 0)
     if arena is None:
         worst = n_updates * (23 + key_len + val_len + 16) + 64
         arena_cap = worst

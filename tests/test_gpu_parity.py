@@ -266,3 +266,46 @@ def test_full_config_shape_parity(olib):
     keys = collect_keys(olib, raw, descs, 6000, per_shard_cap=8)
     check_parity(e, ost, keys, range(0, nshards, 61))
     e.close()
+
+
+def test_snappy_replay_parity(olib):
+    """Config #5: Snappy-compressed payloads, mixed Put/Delete/Merge
+    70/20/10, GPU decompress stage + apply vs oracle decompress + apply."""
+    import ctypes as CT
+    plib = ra.load()
+    nshards, nupd = 64, 20000
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=256, kind=2,
+                                       key_space=1 << 16, seed=77,
+                                       compressible=1)
+    raw = bytes(arena)[:used]
+    # oracle leg: decompress-then-apply must equal plain apply of originals
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd, merge_op=1)
+    # compress each update (host transport side)
+    comp = CT.create_string_buffer(used + used // 4 + 64 * nupd)
+    cdescs = (ra.ffi.GraUpdateDesc * nupd)()
+    ulens = []
+    off = 0
+    for i in range(nupd):
+        d = descs[i]
+        blob = raw[d.off:d.off + d.len]
+        cap = len(blob) + len(blob) // 6 + 64
+        tmp = CT.create_string_buffer(cap)
+        clen = plib.gra_snappy_compress(blob, len(blob), tmp, cap)
+        assert clen > 0
+        CT.memmove(CT.byref(comp, off), tmp, clen)
+        cdescs[i] = ra.ffi.GraUpdateDesc(d.shard, clen, off, d.ts)
+        ulens.append(d.len)
+        off += clen
+    assert off < used  # compressible streams actually compressed
+    e = ra.Engine(nshards=nshards, merge_op=1)
+    rep = e.upload_snappy(CT.cast(comp, CT.POINTER(CT.c_uint8)), off,
+                          cdescs, nupd, ulens)
+    for first in range(0, nupd, 5000):
+        rep.tick(first, 5000)
+    rep.sync()
+    st = e.stats()
+    assert st.snappy_ms > 0
+    keys = collect_keys(olib, raw, descs, 4000)
+    check_parity(e, ost, keys, range(0, nshards, 4))
+    e.close()

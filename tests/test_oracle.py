@@ -215,3 +215,96 @@ def test_randomized_replay_leader_follower_equal(lib):
         assert leader.latest_seq(s) == follower.latest_seq(s)
         for k in keys:
             assert leader.get(s, k) == follower.get(s, k)
+
+
+# ---------- Snappy codec (transport compression, config #5) ----------
+
+def test_snappy_roundtrip_and_crossimpl(lib):
+    """Oracle codec round-trips; product and oracle codecs agree on each
+    other's streams (two independent restatements of the public format)."""
+    import ctypes as CT
+    import rocksplicator_amd as ra
+    plib = ra.load()
+    lib.orc_snappy_compress.restype = CT.c_size_t
+    lib.orc_snappy_compress.argtypes = [CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t]
+    lib.orc_snappy_decompress.argtypes = [CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t, CT.POINTER(CT.c_size_t)]
+    cases = [
+        b"",
+        b"a",
+        b"abcabcabcabcabcabcabcabcabc" * 10,   # short-period copies
+        bytes(range(256)) * 8,                  # periodic
+        random.Random(3).randbytes(5000),       # incompressible
+        (b"x" * 300 + b"y" * 300) * 4,          # long runs
+        b"0123456789abcdef" * 64,               # 16-period (1KB value shape)
+    ]
+    for data in cases:
+        cap = len(data) + len(data) // 6 + 64
+        # oracle compress -> oracle decompress
+        obuf = CT.create_string_buffer(cap)
+        clen = lib.orc_snappy_compress(data, len(data), obuf, cap)
+        assert clen > 0
+        dbuf = CT.create_string_buffer(len(data) + 16)
+        dlen = CT.c_size_t()
+        assert lib.orc_snappy_decompress(obuf.raw[:clen], clen, dbuf, len(data) + 16, CT.byref(dlen)) == 0
+        assert dbuf.raw[:dlen.value] == data
+        # product compress -> oracle decompress (cross-impl)
+        pbuf = CT.create_string_buffer(cap)
+        pclen = plib.gra_snappy_compress(data, len(data), pbuf, cap)
+        assert pclen > 0
+        assert lib.orc_snappy_decompress(pbuf.raw[:pclen], pclen, dbuf, len(data) + 16, CT.byref(dlen)) == 0
+        assert dbuf.raw[:dlen.value] == data
+        # oracle compress -> product decompress (cross-impl)
+        qbuf = CT.create_string_buffer(len(data) + 16)
+        qlen = plib.gra_snappy_decompress(obuf.raw[:clen], clen, qbuf, len(data) + 16)
+        assert qlen == len(data)
+        assert qbuf.raw[:qlen] == data
+
+
+def test_snappy_known_answer(lib):
+    """Hand-derived vectors for the block format."""
+    import ctypes as CT
+    lib.orc_snappy_decompress.argtypes = [CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t, CT.POINTER(CT.c_size_t)]
+    # "abc" as a single literal: varint(3)=03, tag=(3-1)<<2=0x08, "abc"
+    v = bytes([0x03, 0x08]) + b"abc"
+    dbuf = CT.create_string_buffer(32)
+    dlen = CT.c_size_t()
+    assert lib.orc_snappy_decompress(v, len(v), dbuf, 32, CT.byref(dlen)) == 0
+    assert dbuf.raw[:dlen.value] == b"abc"
+    # "aaaaaaaa" via literal "a" + copy(off=1, len=7): varint(8),
+    # tag lit len1 = 0x00, 'a', copy type01: len-4=3 in bits2-4, off=1:
+    # tag = (3<<2)|1 = 0x0D, off-low byte=0x01
+    v = bytes([0x08, 0x00]) + b"a" + bytes([0x0D, 0x01])
+    assert lib.orc_snappy_decompress(v, len(v), dbuf, 32, CT.byref(dlen)) == 0
+    assert dbuf.raw[:dlen.value] == b"a" * 8
+    # 2-byte-offset copy: "abcd"*4 via literal "abcd" + copy(off=4,len=12):
+    # tag type10 = ((12-1)<<2)|2 = 0x2E, off LE = 04 00
+    v = bytes([0x10, 0x0C]) + b"abcd" + bytes([0x2E, 0x04, 0x00])
+    assert lib.orc_snappy_decompress(v, len(v), dbuf, 32, CT.byref(dlen)) == 0
+    assert dbuf.raw[:dlen.value] == b"abcd" * 4
+    # corruption: truncated, bad offset
+    assert lib.orc_snappy_decompress(v[:-1], len(v) - 1, dbuf, 32, CT.byref(dlen)) != 0
+    bad = bytes([0x04, 0x0D, 0x01])  # copy with empty history
+    assert lib.orc_snappy_decompress(bad, len(bad), dbuf, 32, CT.byref(dlen)) != 0
+
+
+def test_compressible_generator_compresses(lib):
+    import rocksplicator_amd as ra
+    plib = ra.load()
+    arena, used, descs = ra.gen_stream(nshards=4, n_updates=200, key_len=16,
+                                       val_len=1024, seed=11, compressible=1)
+    raw = bytes(arena)[:used]
+    import ctypes as CT
+    total_c = 0
+    for i in range(200):
+        d = descs[i]
+        blob = raw[d.off:d.off + d.len]
+        cap = len(blob) + len(blob) // 6 + 64
+        buf = CT.create_string_buffer(cap)
+        clen = plib.gra_snappy_compress(blob, len(blob), buf, cap)
+        assert clen > 0
+        total_c += clen
+        # still a valid WriteBatch after round-trip
+        dbuf = CT.create_string_buffer(len(blob) + 16)
+        qlen = plib.gra_snappy_decompress(buf.raw[:clen], clen, dbuf, len(blob) + 16)
+        assert qlen == len(blob) and dbuf.raw[:qlen] == blob
+    assert total_c < used * 0.5, f"poor compression: {total_c}/{used}"
