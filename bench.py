@@ -50,6 +50,9 @@ def parse_args():
                    help="optional rocprofv3-derived per-launch HBM traffic (see profiles/)")
     p.add_argument("--h2d", action="store_true",
                    help="measure the PCIe-inclusive staging path instead (side report)")
+    p.add_argument("--snappy", action="store_true",
+                   help="BASELINE config #5: Snappy-compressed payloads, "
+                        "GPU decompress stage ahead of the decode walk")
     p.add_argument("--repartition", action="store_true",
                    help="BASELINE config #4: re-home update blobs to their "
                         "owner rank via RCCL all-to-all over xGMI inside "
@@ -65,7 +68,7 @@ def dist_env():
     return rank, world, local
 
 
-def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s):
+def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s, snappy=False):
     """Time the ORACLE applier (the CPU restatement of replicated_db.cpp:
     369-383 + rocksdb_wrapper.cpp:13-28) on the host cores — checker/baseline
     only, never the measured GPU path."""
@@ -78,10 +81,13 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s):
     ods = (oracle_ffi.OrcUpdateDesc * n)()
     C.memmove(ods, descs, C.sizeof(oracle_ffi.OrcUpdateDesc) * n)
 
+    bench_fn = (olib.orc_cpu_snappy_apply_bench if snappy
+                else olib.orc_cpu_apply_bench)
+
     def run(sample_n, threads):
         st = olib.orc_store_create(nshards, 0)
         try:
-            secs = olib.orc_cpu_apply_bench(
+            secs = bench_fn(
                 st, C.cast(raw_arena, C.c_void_p), ods, sample_n, threads)
         finally:
             olib.orc_store_destroy(st)
@@ -101,8 +107,9 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s):
         "unit": "updates/s",
         "cores": ncores,
         "kind": "port",
-        "sample": f"{reps}x{sample_n} updates of the same synthetic stream, "
-                  f"{ncores} threads, {total_s:.2f}s total",
+        "sample": f"{reps}x{sample_n} updates of the same synthetic stream"
+                  + (" (snappy decompress+apply)" if snappy else "") +
+                  f", {ncores} threads, {total_s:.2f}s total",
     }
 
 
@@ -228,7 +235,8 @@ def main():
     arena_p = C.cast(arena, C.POINTER(C.c_uint8))
     descs = (ra.ffi.GraUpdateDesc * n_upd)()
     g = ra.ffi.GraGenOpts(args.nshards, args.key_len, args.val_len, args.kind,
-                          1 << 24, 0.99, 0xB0CC5EED + 1000 * rank)
+                          1 << 24, 0.99, 0xB0CC5EED + 1000 * rank,
+                          1 if args.snappy else 0, 0)
     used = C.c_size_t()
     rc = ra.load().gra_gen_stream(C.byref(g), n_upd, arena_p, worst,
                                   C.byref(used), descs, 0)
@@ -240,7 +248,27 @@ def main():
     store_bytes = min(int(used * 1.3) + (1 << 30), 48 << 30)
     eng = ra.Engine(nshards=args.nshards, device=local, store_ring=1,
                     store_bytes=store_bytes)
-    if args.h2d:
+    if args.snappy:
+        # transport compression (config #5): compress on host (untimed),
+        # the GPU decompress stage runs inside every timed tick
+        lib = ra.load()
+        comp_cap = used + used // 4 + 64 * n_upd
+        comp = (C.c_uint8 * comp_cap)()
+        cdescs = (ra.ffi.GraUpdateDesc * n_upd)()
+        ulens = (C.c_uint32 * n_upd)()
+        comp_used = C.c_size_t()
+        rc = lib.gra_snappy_compress_stream(
+            arena_p, descs, n_upd, C.cast(comp, C.POINTER(C.c_uint8)),
+            comp_cap, C.byref(comp_used), cdescs, ulens, os.cpu_count() or 8)
+        assert rc == 0, f"compress rc={rc}"
+        counts = (C.c_uint32 * n_upd)(*([1] * n_upd))
+        out = C.c_void_p()
+        rc = lib.gra_upload_snappy(eng.h, C.cast(comp, C.POINTER(C.c_uint8)),
+                                   comp_used.value, cdescs, n_upd, ulens,
+                                   counts, C.byref(out))
+        assert rc == 0, ra.ffi.last_error(lib)
+        rep = ra.ffi.Replay(eng, out)
+    elif args.h2d:
         pin = eng.pin_alloc(used)
         C.memmove(pin, arena, used)
         rep = eng.upload(pin, used, descs, n_upd)
@@ -321,7 +349,8 @@ def main():
         "config": {
             "workload": f"{args.nshards}shards_{args.key_len}Bkey_"
                         f"{args.val_len}Bval_" +
-                        {0: "uniform", 1: "zipf0.99", 2: "mixed70/20/10"}[args.kind],
+                        {0: "uniform", 1: "zipf0.99", 2: "mixed70/20/10"}[args.kind]
+                        + ("_snappy" if args.snappy else ""),
             "nshards": args.nshards,
             "key_len": args.key_len,
             "val_len": args.val_len,
@@ -339,6 +368,7 @@ def main():
         },
         "kernels_ms_per_tick": {
             "h2d": st.h2d_ms / launches,
+            "snappy": st.snappy_ms / launches,
             "decode": st.decode_ms / launches,
             "scan": st.scan_ms / launches,
             "emit": st.emit_ms / launches,
@@ -351,8 +381,13 @@ def main():
     }
 
     if rank == 0 and args.cpu_baseline:
-        result["cpu_baseline"] = cpu_baseline_leg(
-            arena, used, descs, n_upd, args.nshards, args.cpu_sample_seconds)
+        if args.snappy:
+            result["cpu_baseline"] = cpu_baseline_leg(
+                comp, comp_used.value, cdescs, n_upd, args.nshards,
+                args.cpu_sample_seconds, snappy=True)
+        else:
+            result["cpu_baseline"] = cpu_baseline_leg(
+                arena, used, descs, n_upd, args.nshards, args.cpu_sample_seconds)
         if result["cpu_baseline"]["value"]:
             result["gpu_vs_cpu"] = value / result["cpu_baseline"]["value"]
 

@@ -178,3 +178,40 @@ extern "C" uint32_t gra_snappy_decompress(const uint8_t *src, uint32_t slen,
                                           uint8_t *dst, uint32_t dcap) {
   return snp::decompress(src, slen, dst, dcap);
 }
+
+#include <thread>
+
+/* Parallel host compression of a generated stream (harness transport side).
+ * Slot layout uses worst-case per-update offsets so threads write
+ * independently; out_descs carry the real compressed lengths. */
+extern "C" int gra_snappy_compress_stream(const uint8_t *arena,
+                                          const GraUpdateDesc *descs,
+                                          uint64_t n, uint8_t *out,
+                                          size_t out_cap, size_t *out_used,
+                                          GraUpdateDesc *out_descs,
+                                          uint32_t *ulens, int nthreads) {
+  std::vector<uint64_t> off(n + 1);
+  off[0] = 0;
+  for (uint64_t i = 0; i < n; i++)
+    off[i + 1] = off[i] + snp::max_compressed_len(descs[i].len);
+  if (off[n] > out_cap) return GRA_FULL;
+  if (nthreads < 1) nthreads = 1;
+  std::vector<std::thread> ths;
+  for (int t = 0; t < nthreads; t++) {
+    ths.emplace_back([&, t] {
+      for (uint64_t i = t; i < n; i += (uint64_t)nthreads) {
+        const GraUpdateDesc &d = descs[i];
+        uint32_t clen = snp::compress(arena + d.off, d.len, out + off[i],
+                                      (uint32_t)(off[i + 1] - off[i]));
+        out_descs[i].shard = d.shard;
+        out_descs[i].len = clen;
+        out_descs[i].off = off[i];
+        out_descs[i].ts = d.ts;
+        ulens[i] = d.len;
+      }
+    });
+  }
+  for (auto &th : ths) th.join();
+  if (out_used) *out_used = off[n];
+  return GRA_OK;
+}

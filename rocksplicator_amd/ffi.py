@@ -130,6 +130,10 @@ def load():
     lib.gra_snappy_compress.argtypes = [C.c_char_p, C.c_uint32, C.c_char_p, C.c_uint32]
     lib.gra_snappy_decompress.restype = C.c_uint32
     lib.gra_snappy_decompress.argtypes = [C.c_char_p, C.c_uint32, C.c_char_p, C.c_uint32]
+    lib.gra_snappy_compress_stream.argtypes = [
+        C.POINTER(C.c_uint8), C.POINTER(GraUpdateDesc), C.c_uint64,
+        C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(C.c_size_t),
+        C.POINTER(GraUpdateDesc), C.POINTER(C.c_uint32), C.c_int]
     _lib = lib
     return lib
 
