@@ -40,8 +40,8 @@ def parse_args():
     p.add_argument("--key-len", type=int, default=16)
     p.add_argument("--val-len", type=int, default=1024)
     p.add_argument("--kind", type=int, default=1, help="0 uniform,1 zipf,2 mixed")
-    p.add_argument("--tick-updates", type=int, default=204800)
-    p.add_argument("--max-ticks-resident", type=int, default=12,
+    p.add_argument("--tick-updates", type=int, default=819200)
+    p.add_argument("--max-ticks-resident", type=int, default=6,
                    help="distinct ticks of data generated/uploaded; steps cycle over them")
     p.add_argument("--cpu-baseline", action="store_true", default=True)
     p.add_argument("--no-cpu-baseline", dest="cpu_baseline", action="store_false")
