@@ -1110,7 +1110,7 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
     udescs[i].len = ulens[i];
     udescs[i].off = scratch;
     udescs[i].ts = descs[i].ts;
-    scratch += ((uint64_t)ulens[i] + 8 + 15) & ~15ULL;
+    scratch += ((uint64_t)ulens[i] + 16 + 15) & ~15ULL; /* 16B chunk slack */
   }
   r->arena_bytes = scratch;
   if (hipMalloc(&r->d_blobs, scratch + 16) != hipSuccess ||

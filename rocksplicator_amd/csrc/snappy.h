@@ -23,9 +23,9 @@ namespace snp {
 WB_HD uint32_t max_compressed_len(uint32_t n) { return 32 + n + n / 6; }
 
 /* Decompress src[0..slen) into dst (capacity dcap). Returns uncompressed
- * length, or UINT32_MAX on corruption. dst regions may be written in 8-byte
- * chunks up to 7 bytes past the uncompressed length — callers provide
- * >= 8 bytes of slack per output slot. */
+ * length, or UINT32_MAX on corruption. dst regions may be written in wide
+ * chunks up to 15 bytes past the uncompressed length — callers provide
+ * >= 16 bytes of slack per output slot. */
 WB_HD uint32_t decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
                           uint32_t dcap) {
   uint32_t ulen = 0;
@@ -45,13 +45,13 @@ WB_HD uint32_t decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
         ip += nb;
       }
       if (ip + len > slen || op + len > ulen) return UINT32_MAX;
-      /* literals never overlap: 8-byte chunks (arena + slot slack cover the
-       * over-read/over-write) */
-      for (uint32_t b = 0; b < len; b += 8) {
+      /* literals never overlap: 16-byte chunks (arena + slot slack cover
+       * the over-read/over-write; gfx950 tolerates misaligned dwordx4) */
+      for (uint32_t b = 0; b < len; b += 16) {
 #if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
-        *(uint64_t *)(dst + op + b) = *(const uint64_t *)(src + ip + b);
+        *(uint4 *)(dst + op + b) = *(const uint4 *)(src + ip + b);
 #else
-        for (uint32_t j = b; j < b + 8 && j < len; j++) dst[op + j] = src[ip + j];
+        for (uint32_t j = b; j < b + 16 && j < len; j++) dst[op + j] = src[ip + j];
 #endif
       }
       ip += len;
@@ -76,7 +76,16 @@ WB_HD uint32_t decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
         ip += 4;
       }
       if (off == 0 || off > op || op + len > ulen) return UINT32_MAX;
-      if (off >= 8) { /* chunked forward copy is overlap-safe at off>=8 */
+      if (off >= 16) { /* chunked forward copy is overlap-safe at off>=chunk */
+        for (uint32_t b = 0; b < len; b += 16) {
+#if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
+          *(uint4 *)(dst + op + b) = *(const uint4 *)(dst + op - off + b);
+#else
+          for (uint32_t j = b; j < b + 16 && j < len; j++)
+            dst[op + j] = dst[op + j - off];
+#endif
+        }
+      } else if (off >= 8) {
         for (uint32_t b = 0; b < len; b += 8) {
 #if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
           *(uint64_t *)(dst + op + b) = *(const uint64_t *)(dst + op - off + b);
