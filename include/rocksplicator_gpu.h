@@ -55,6 +55,12 @@ typedef struct GraEngineOpts {
   uint64_t store_bytes;    /* device run-arena capacity (0 = 4 GiB default) */
   uint64_t staging_bytes;  /* pinned staging per buffer (0 = 256 MiB default) */
   uint32_t max_wb_records; /* per-batch record cap (0 = 1024 default) */
+  int retain_log;          /* 1: retain applied/written rep blobs per shard so
+                              this node can serve downstream pulls (the WAL-
+                              retention analog; replicated_db.cpp:435-575) */
+  uint64_t log_bytes;      /* retained-log cap across shards (0 = 256 MiB);
+                              oldest batches evicted first (WAL_ttl analog,
+                              performance.cpp:99) */
 } GraEngineOpts;
 
 void gra_engine_opts_init(GraEngineOpts *opts); /* fill defaults */
@@ -103,6 +109,24 @@ int gra_get(GraDb *db, const void *key, size_t klen, void *buf, size_t cap,
 /* Drain barrier: every update submitted so far is applied and durable; after
  * this, gra_latest_seq and gra_get are linearizable per shard. */
 int gra_flush(GraEngine *e);
+
+/* ---------------- leader update-serving ----------------
+ * ≅ ReplicatedDB::handleReplicateRequest batch read-out (replicated_db.cpp:
+ * 435-575): serve retained batches with base seq > since_seq, up to
+ * max_updates (the reference default is 50, replicated_db.cpp:42-43), as
+ * (seq, timestamp, rep bytes) triples — the Update wire triple
+ * (replicator.thrift:44-57). Blobs are copied into buf; out[i].off/len
+ * index it. Requires opts.retain_log. Returns GRA_OK with *n_out = 0 when
+ * caught up; GRA_ERR when since_seq predates the retained log (the
+ * reference's WAL-gone case). */
+typedef struct {
+  uint64_t seq;   /* batch base seq (first record's seq) */
+  int64_t ts;     /* timestamp the leader stamped (ms) */
+  uint32_t off, len;
+} GraServedUpdate;
+int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
+                    GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
+                    size_t cap);
 
 /* ---------------- WriteBatch builder ----------------
  * The reference's callers construct updates with rocksdb::WriteBatch

@@ -31,6 +31,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <chrono>
 #include <cstdio>
 #include <cstring>
 #include <deque>
@@ -891,11 +892,29 @@ GraDb *gra_open(GraEngine *e, uint32_t shard) {
 }
 void gra_close(GraDb *db) { delete db; }
 
+/* Append a batch to the shard's retained update log (under ss.mu), evicting
+ * oldest entries past the per-shard byte budget — the WAL-retention analog
+ * (bounded like WAL_ttl_seconds bounds the reference's serving window). */
+static void retain_locked(GraEngine *e, ShardState &ss, uint64_t base_seq,
+                          uint32_t count, int64_t ts, const uint8_t *rep,
+                          size_t len) {
+  uint64_t per_shard = (e->opts.log_bytes ? e->opts.log_bytes : 256ULL << 20) /
+                       (e->opts.nshards ? e->opts.nshards : 1);
+  LogEnt ent;
+  ent.base_seq = base_seq;
+  ent.count = count;
+  ent.ts = ts;
+  ent.rep.assign(rep, rep + len);
+  ss.log_used += len;
+  ss.log.push_back(std::move(ent));
+  while (ss.log_used > per_shard && ss.log.size() > 1) {
+    ss.log_used -= ss.log.front().rep.size();
+    ss.log.pop_front();
+  }
+}
+
 int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
                                   int64_t ts) {
-  (void)ts; /* latency metric only; the LogData(ts) trailer the reference
-             * appends (rocksdb_wrapper.cpp:19-20) is WAL-only and never
-             * reaches the memtable */
   GraEngine *e = db->e;
   ShardState &ss = e->shards[db->shard];
   uint64_t base;
@@ -911,6 +930,10 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     count = wb::fixed32_le(rep + 8);
     base = ss.next_seq;
     ss.next_seq += count;
+    /* ts beyond this point is wire metadata only: the LogData(ts) trailer
+     * the reference appends (rocksdb_wrapper.cpp:19-20) is WAL-only and
+     * never reaches the memtable. Chained nodes re-serve with this ts. */
+    if (e->opts.retain_log) retain_locked(e, ss, base, count, ts, rep, len);
   }
   std::lock_guard<std::mutex> lk(e->mu);
   if (e->stage_fill + len + 16 > e->opts.staging_bytes ||
@@ -938,10 +961,12 @@ uint64_t gra_latest_seq(GraDb *db) {
 
 int gra_write_leader(GraDb *db, const uint8_t *rep, size_t len,
                      uint64_t *seq_out) {
-  ShardState &ss = db->e->shards[db->shard];
+  GraEngine *e = db->e;
+  ShardState &ss = e->shards[db->shard];
   std::lock_guard<std::mutex> lk(ss.mu);
   auto run = std::make_shared<Run>();
-  if (!host_build_run(rep, len, ss.next_seq, run.get())) {
+  uint64_t base = ss.next_seq;
+  if (!host_build_run(rep, len, base, run.get())) {
     g_err = "corrupt WriteBatch rep";
     return GRA_CORRUPT;
   }
@@ -949,7 +974,46 @@ int gra_write_leader(GraDb *db, const uint8_t *rep, size_t len,
   ss.next_seq += count;
   ss.durable_seq = ss.next_seq - 1;
   if (run->n_entries > 0) ss.runs.push_back(std::move(run));
+  if (e->opts.retain_log) {
+    /* the leader stamps writes with now-ms (replicated_db.cpp:115-117's
+     * PutLogData breadcrumb); serving returns it as Update.timestamp */
+    int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
+        std::chrono::system_clock::now().time_since_epoch()).count());
+    retain_locked(e, ss, base, count, now_ms, rep, len);
+  }
   if (seq_out) *seq_out = ss.durable_seq;
+  return GRA_OK;
+}
+
+int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
+                    GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
+                    size_t cap) {
+  GraEngine *e = db->e;
+  if (!e->opts.retain_log) {
+    g_err = "retain_log disabled on this engine";
+    return GRA_ERR;
+  }
+  ShardState &ss = e->shards[db->shard];
+  std::lock_guard<std::mutex> lk(ss.mu);
+  *n_out = 0;
+  if (!ss.log.empty() && since_seq + 1 < ss.log.front().base_seq) {
+    /* reference analog: WAL no longer reaches back that far */
+    g_err = "retained log truncated before requested seq";
+    return GRA_ERR;
+  }
+  size_t off = 0;
+  for (const LogEnt &ent : ss.log) {
+    if (*n_out >= max_updates) break;
+    if (ent.base_seq <= since_seq) continue;
+    if (off + ent.rep.size() > cap) break;
+    memcpy(buf + off, ent.rep.data(), ent.rep.size());
+    out[*n_out].seq = ent.base_seq;
+    out[*n_out].ts = ent.ts;
+    out[*n_out].off = (uint32_t)off;
+    out[*n_out].len = (uint32_t)ent.rep.size();
+    off += ent.rep.size();
+    (*n_out)++;
+  }
   return GRA_OK;
 }
 

@@ -15,6 +15,7 @@
 #pragma once
 #include <cstdint>
 #include <cstring>
+#include <deque>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -38,12 +39,23 @@ struct Run {
   bool resident() const { return !hdrs.empty() || n_entries == 0; }
 };
 
+/* One retained batch for downstream serving (leader update log — the WAL
+ * retention analog; replicated_db.cpp:435-575). */
+struct LogEnt {
+  uint64_t base_seq = 0;
+  uint32_t count = 0;
+  int64_t ts = 0;
+  std::vector<uint8_t> rep;
+};
+
 struct ShardState {
   mutable std::mutex mu;
   uint64_t durable_seq = 0;   /* last seq applied & synced on device */
   uint64_t next_seq = 1;      /* next seq to assign at submission */
   bool poisoned = false;      /* corrupt batch seen; next HRR returns false */
   std::vector<std::shared_ptr<Run>> runs; /* oldest .. newest */
+  std::deque<LogEnt> log;     /* retained batches (retain_log mode) */
+  uint64_t log_used = 0;      /* bytes retained in this shard's log */
 };
 
 /* Get over a run list (newest last). merge_op: 0 concat, 1 u64add.

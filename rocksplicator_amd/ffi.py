@@ -32,6 +32,17 @@ class GraEngineOpts(C.Structure):
         ("store_bytes", C.c_uint64),
         ("staging_bytes", C.c_uint64),
         ("max_wb_records", C.c_uint32),
+        ("retain_log", C.c_int),
+        ("log_bytes", C.c_uint64),
+    ]
+
+
+class GraServedUpdate(C.Structure):
+    _fields_ = [
+        ("seq", C.c_uint64),
+        ("ts", C.c_int64),
+        ("off", C.c_uint32),
+        ("len", C.c_uint32),
     ]
 
 
@@ -100,6 +111,10 @@ def load():
     lib.gra_write_leader.argtypes = [C.c_void_p, C.c_char_p, C.c_size_t, C.POINTER(C.c_uint64)]
     lib.gra_get.argtypes = [C.c_void_p, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t, C.POINTER(C.c_size_t)]
     lib.gra_flush.argtypes = [C.c_void_p]
+    lib.gra_get_updates.argtypes = [C.c_void_p, C.c_uint64, C.c_uint32,
+                                    C.POINTER(GraServedUpdate),
+                                    C.POINTER(C.c_uint32), C.c_char_p,
+                                    C.c_size_t]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
     lib.gra_pin_free.argtypes = [C.c_void_p, C.POINTER(C.c_uint8)]
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
@@ -197,7 +212,7 @@ class Engine:
     fallback exists for the follower apply path."""
 
     def __init__(self, nshards, device=-1, merge_op=0, store_ring=0,
-                 store_bytes=0, staging_bytes=0):
+                 store_bytes=0, staging_bytes=0, retain_log=0, log_bytes=0):
         self.lib = load()
         opts = GraEngineOpts()
         self.lib.gra_engine_opts_init(C.byref(opts))
@@ -205,6 +220,9 @@ class Engine:
         opts.device = device
         opts.merge_op = merge_op
         opts.store_ring = store_ring
+        opts.retain_log = retain_log
+        if log_bytes:
+            opts.log_bytes = log_bytes
         if store_bytes:
             opts.store_bytes = store_bytes
         if staging_bytes:
@@ -302,6 +320,19 @@ class Db:
         if rc != GRA_OK:
             raise RuntimeError(f"gra_get rc={rc}: {last_error(self.lib)}")
         return buf.raw[: vlen.value]
+
+    def get_updates(self, since_seq, max_updates=50, cap=1 << 22):
+        """Leader serving: [(seq, ts, rep_bytes), ...] with
+        base seq > since_seq — the reference Update triple."""
+        out = (GraServedUpdate * max_updates)()
+        buf = C.create_string_buffer(cap)
+        n = C.c_uint32()
+        rc = self.lib.gra_get_updates(self.h, since_seq, max_updates, out,
+                                      C.byref(n), buf, cap)
+        if rc != GRA_OK:
+            raise RuntimeError(f"gra_get_updates rc={rc}: {last_error(self.lib)}")
+        return [(out[i].seq, out[i].ts, buf.raw[out[i].off:out[i].off + out[i].len])
+                for i in range(n.value)]
 
 
 class Replay:
