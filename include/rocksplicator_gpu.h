@@ -128,6 +128,14 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
                     GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
                     size_t cap);
 
+/* ≅ MaxNumberBox::wait (max_number_box.h:38-83, .cpp:63) behind the 2-ACK
+ * write modes (replicated_db.cpp:147-156): block until the downstream ack
+ * reaches seq. confirmed=1 waits for follower-applied progress (mode 2 —
+ * the pull request's seq_no, replicated_db.cpp:452-456); confirmed=0 waits
+ * for served progress (mode 1, :543-546). GRA_OK, or GRA_NOT_FOUND on
+ * timeout. */
+int gra_wait_ack(GraDb *db, uint64_t seq, int confirmed, int timeout_ms);
+
 /* ---------------- WriteBatch builder ----------------
  * The reference's callers construct updates with rocksdb::WriteBatch
  * (e.g. examples/counter_service/counter_handler.cpp:152-158); this is the

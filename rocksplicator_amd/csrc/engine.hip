@@ -995,6 +995,9 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
   }
   ShardState &ss = e->shards[db->shard];
   std::lock_guard<std::mutex> lk(ss.mu);
+  /* the pull request's seq_no IS the follower's confirmed progress
+   * (mode-2 ack, replicated_db.cpp:452-456) */
+  if (since_seq > ss.acked_confirmed) ss.acked_confirmed = since_seq;
   *n_out = 0;
   if (!ss.log.empty() && since_seq + 1 < ss.log.front().base_seq) {
     /* reference analog: WAL no longer reaches back that far */
@@ -1014,7 +1017,31 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
     off += ent.rep.size();
     (*n_out)++;
   }
+  if (*n_out > 0) { /* mode-1 ack: acked once sent (replicated_db.cpp:543-546) */
+    uint64_t last = out[*n_out - 1].seq;
+    const LogEnt *le = nullptr;
+    for (const LogEnt &ent : ss.log)
+      if (ent.base_seq == last) le = &ent;
+    uint64_t sent = le ? le->base_seq + le->count - 1 : last;
+    if (sent > ss.acked_sent) ss.acked_sent = sent;
+  }
+  ss.ack_cv.notify_all();
   return GRA_OK;
+}
+
+/* ≅ MaxNumberBox::wait (max_number_box.cpp:63) behind ReplicatedDB::Write's
+ * 2-ACK modes (replicated_db.cpp:147-156): block until the downstream ack
+ * (confirmed: the follower applied it; else: it was served) reaches seq, or
+ * timeout. Returns GRA_OK / GRA_NOT_FOUND on timeout. */
+int gra_wait_ack(GraDb *db, uint64_t seq, int confirmed, int timeout_ms) {
+  ShardState &ss = db->e->shards[db->shard];
+  std::unique_lock<std::mutex> lk(ss.mu);
+  auto reached = [&] {
+    return (confirmed ? ss.acked_confirmed : ss.acked_sent) >= seq;
+  };
+  if (ss.ack_cv.wait_for(lk, std::chrono::milliseconds(timeout_ms), reached))
+    return GRA_OK;
+  return GRA_NOT_FOUND;
 }
 
 int gra_flush(GraEngine *e) {

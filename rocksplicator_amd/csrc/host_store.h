@@ -13,6 +13,7 @@
  * its seq.
  */
 #pragma once
+#include <condition_variable>
 #include <cstdint>
 #include <cstring>
 #include <deque>
@@ -56,6 +57,12 @@ struct ShardState {
   std::vector<std::shared_ptr<Run>> runs; /* oldest .. newest */
   std::deque<LogEnt> log;     /* retained batches (retain_log mode) */
   uint64_t log_used = 0;      /* bytes retained in this shard's log */
+  /* follower-ACK box ≅ MaxNumberBox (max_number_box.h:38-83): serving a
+   * pull carries the follower's progress — the request's seq_no is the
+   * confirmed ack (mode 2, replicated_db.cpp:452-456), and everything sent
+   * is the sent ack (mode 1, :543-546). */
+  uint64_t acked_sent = 0, acked_confirmed = 0;
+  std::condition_variable ack_cv;
 };
 
 /* Get over a run list (newest last). merge_op: 0 concat, 1 u64add.

@@ -115,6 +115,7 @@ def load():
                                     C.POINTER(GraServedUpdate),
                                     C.POINTER(C.c_uint32), C.c_char_p,
                                     C.c_size_t]
+    lib.gra_wait_ack.argtypes = [C.c_void_p, C.c_uint64, C.c_int, C.c_int]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
     lib.gra_pin_free.argtypes = [C.c_void_p, C.POINTER(C.c_uint8)]
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
@@ -320,6 +321,12 @@ class Db:
         if rc != GRA_OK:
             raise RuntimeError(f"gra_get rc={rc}: {last_error(self.lib)}")
         return buf.raw[: vlen.value]
+
+    def wait_ack(self, seq, confirmed=True, timeout_ms=2000):
+        """MaxNumberBox::wait equivalent: True when the downstream ack
+        reached seq before the timeout."""
+        return self.lib.gra_wait_ack(self.h, seq, 1 if confirmed else 0,
+                                     timeout_ms) == GRA_OK
 
     def get_updates(self, since_seq, max_updates=50, cap=1 << 22):
         """Leader serving: [(seq, ts, rep_bytes), ...] with

@@ -122,3 +122,68 @@ def test_log_truncation_detected():
     ups = ldb.get_updates(60)
     assert len(ups) == 4 and ups[0][0] == 61
     leader.close()
+
+
+def test_replicator_registry_modes_and_roles(olib):
+    """Replicator registry (≅ RocksDBReplicator): background pull threads,
+    2-ACK write mode blocking until the follower applied (mode 2), role
+    transitions, WRITE_TO_SLAVE, and removeDB."""
+    from rocksplicator_amd.replicator import (FOLLOWER, LEADER, Replicator,
+                                              WriteToSlaveError)
+    leader_e = ra.Engine(nshards=16, merge_op=1, retain_log=1)
+    follower_e = ra.Engine(nshards=16, merge_op=1, retain_log=1)
+    lrep = Replicator(leader_e)
+    frep = Replicator(follower_e)
+    lrs = lrep.add_db("db0", LEADER)
+    frep.add_db("db0", FOLLOWER, upstream_db=lrs.db)
+    one = (1).to_bytes(8, "little")
+    # mode 2: returns only after the follower pulled AND applied
+    for i in range(30):
+        seq = lrep.write("db0", PyBatch().merge(b"ctr", one).data(), mode=2)
+        assert seq == i + 1
+    fdb = frep.get("db0").db
+    assert fdb.latest_seq() == 30
+    assert int.from_bytes(fdb.get(b"ctr"), "little") == 30
+    # writes to a follower throw (ReturnCode::WRITE_TO_SLAVE)
+    with pytest.raises(WriteToSlaveError):
+        frep.write("db0", PyBatch().put(b"x", b"y").data())
+    # promote the follower; it serves and accepts writes (role transition)
+    frep.change_role("db0", LEADER)
+    frep.write("db0", PyBatch().merge(b"ctr", one).data())
+    frep.engine.flush()
+    assert int.from_bytes(fdb.get(b"ctr"), "little") == 31
+    # mode-1 ack (served) succeeds once a downstream pulls; without any
+    # downstream the mode-1 write times out but still commits locally
+    seq = frep.write("db0", PyBatch().merge(b"ctr", one).data(), mode=1)
+    frep.engine.flush()
+    assert frep.get("db0").db.latest_seq() == seq
+    lrep.remove_db("db0")
+    lrep.close()
+    frep.close()
+    leader_e.close()
+    follower_e.close()
+
+
+def test_write_degradation_counter():
+    """No follower attached: mode-2 writes miss their ACK; after the miss
+    threshold the wait degrades to the short timeout
+    (replicated_db.cpp:236-273 behavior)."""
+    import time as _t
+    from rocksplicator_amd.replicator import LEADER, Replicator
+    e = ra.Engine(nshards=4, retain_log=1)
+    rep = Replicator(e)
+    rep.ACK_TIMEOUT_MS = 60
+    rep.DEGRADE_AFTER_MISSES = 3
+    rep.add_db("d", LEADER)
+    blob = PyBatch().put(b"k", b"v").data()
+    t0 = _t.time()
+    for _ in range(3):
+        rep.write("d", blob, mode=2)  # 3 full-timeout misses
+    mid = _t.time()
+    for _ in range(5):
+        rep.write("d", blob, mode=2)  # degraded: ~10ms each
+    t1 = _t.time()
+    assert mid - t0 >= 0.15  # 3 x 60ms
+    assert t1 - mid < 0.15   # 5 x ~10ms degraded
+    rep.close()
+    e.close()
