@@ -114,12 +114,14 @@ __global__ void k_snappy(const uint8_t *__restrict__ comp,
                          const UpdDesc *__restrict__ descs, uint32_t n,
                          uint8_t *__restrict__ scratch,
                          uint32_t *__restrict__ err_ring, uint32_t tick) {
-  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t lane = threadIdx.x & 15u;
+  uint32_t i = (blockIdx.x * blockDim.x + threadIdx.x) >> 4;
   if (i >= n) return;
   SnapTask t = tasks[i];
   uint8_t *dst = scratch + descs[i].off;
-  uint32_t r = snp::decompress(comp + t.comp_off, t.comp_len, dst, t.ulen);
-  if (r != t.ulen) {
+  uint32_t r = snp::decompress_coop16(comp + t.comp_off, t.comp_len, dst,
+                                      t.ulen, lane);
+  if (r != t.ulen && lane == 0) {
     for (int b = 0; b < 13; b++) dst[b] = 0xFF; /* force decode rejection */
     atomicAdd(&err_ring[tick % 64u], 1u);
   }
@@ -671,7 +673,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   HIP_TRY(hipEventRecord(t.ev[1], stream)); /* after h2d */
   uint32_t nb = (n + 255) / 256;
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena */
-    hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, stream, d_comp,
+    uint32_t nb16 = (n * 16 + 255) / 256; /* 16 cooperative lanes per update */
+    hipLaunchKernelGGL(k_snappy, dim3(nb16), dim3(256), 0, stream, d_comp,
                        d_snaptasks, d_descw, n, (uint8_t *)d_blobs, d_err_ring,
                        tick);
     HIP_TRY(hipGetLastError());

@@ -169,3 +169,75 @@ inline uint32_t compress(const uint8_t *src, uint32_t slen, uint8_t *dst,
   return o;
 }
 } /* namespace snp */
+
+#ifdef __HIP__
+namespace snp {
+
+/* Cooperative decompress: 16 lanes share ONE stream. Every lane parses the
+ * element headers redundantly (identical data -> converged control flow
+ * within the group; cross-group divergence in a wave drops from 64-way to
+ * 4-way), and the element payload moves with 16 parallel 16-byte chunks
+ * (coalesced within the output slot) instead of one lane's serial loop.
+ * Overlapping copies (off < len) read only bytes before the element start
+ * via the period-off pattern, so chunk order doesn't matter.
+ * Same format/validation as snp::decompress; same >=16 B slot slack. */
+__device__ inline uint32_t decompress_coop16(const uint8_t *__restrict__ src,
+                                             uint32_t slen,
+                                             uint8_t *__restrict__ dst,
+                                             uint32_t dcap, uint32_t lane) {
+  uint32_t ulen = 0;
+  uint32_t ip = wb::varint32(src, slen, &ulen);
+  if (ip == 0 || ulen > dcap) return UINT32_MAX;
+  uint32_t op = 0;
+  while (ip < slen) {
+    uint8_t tag = src[ip++];
+    if ((tag & 3) == 0) { /* literal */
+      uint32_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        uint32_t nb = len - 60;
+        if (ip + nb > slen) return UINT32_MAX;
+        len = 0;
+        for (uint32_t b = 0; b < nb; b++) len |= (uint32_t)src[ip + b] << (8 * b);
+        len += 1;
+        ip += nb;
+      }
+      if (ip + len > slen || op + len > ulen) return UINT32_MAX;
+      for (uint32_t b = lane * 16; b < len; b += 16 * 16)
+        *(uint4 *)(dst + op + b) = *(const uint4 *)(src + ip + b);
+      ip += len;
+      op += len;
+    } else {
+      uint32_t len, off;
+      if ((tag & 3) == 1) {
+        len = ((tag >> 2) & 7) + 4;
+        if (ip + 1 > slen) return UINT32_MAX;
+        off = ((uint32_t)(tag >> 5) << 8) | src[ip];
+        ip += 1;
+      } else if ((tag & 3) == 2) {
+        len = (tag >> 2) + 1;
+        if (ip + 2 > slen) return UINT32_MAX;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+      } else {
+        len = (tag >> 2) + 1;
+        if (ip + 4 > slen) return UINT32_MAX;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+              ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > op || op + len > ulen) return UINT32_MAX;
+      if (off >= len) { /* no overlap: parallel 16B chunks */
+        for (uint32_t b = lane * 16; b < len; b += 16 * 16)
+          *(uint4 *)(dst + op + b) = *(const uint4 *)(dst + op - off + b);
+      } else { /* periodic pattern: read only pre-element bytes */
+        for (uint32_t j = lane; j < len; j += 16)
+          dst[op + j] = dst[op - off + (j % off)];
+      }
+      op += len;
+    }
+  }
+  return op == ulen ? op : UINT32_MAX;
+}
+
+} /* namespace snp */
+#endif /* __HIP__ */
