@@ -128,6 +128,17 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
                     GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
                     size_t cap);
 
+/* Per-db counters ≅ the replicator stats hooks' per-db fan-out
+ * (rocksdb_replicator/replicator_stats.cpp:33-102: replicator_in_bytes,
+ * replicator_out_bytes, replicator_latency_ms, handle-response failures). */
+typedef struct {
+  uint64_t updates_applied, in_bytes, apply_failures;
+  uint64_t updates_served, out_bytes;
+  uint64_t latency_ms_sum, latency_samples;
+  uint64_t latest_seq;
+} GraDbCounters;
+int gra_db_counters(GraDb *db, GraDbCounters *out);
+
 /* ≅ MaxNumberBox::wait (max_number_box.h:38-83, .cpp:63) behind the 2-ACK
  * write modes (replicated_db.cpp:147-156): block until the downstream ack
  * reaches seq. confirmed=1 waits for follower-applied progress (mode 2 —

@@ -419,7 +419,6 @@ struct Slot {
   DevRunDesc *h_rundescs = nullptr;
   DevRunDesc *d_rundescs = nullptr; /* per-slot device buffer so the D2H can
                                        overlap the next tick's kernels */
-  uint16_t *h_recn = nullptr;
   uint32_t *h_err = nullptr;
   UpdDesc *h_descs = nullptr;      /* staging-path desc upload */
   bool busy = false;
@@ -567,7 +566,6 @@ int GraEngine::init(const GraEngineOpts &o) {
     HIP_TRY(hipHostMalloc(&s.h_groups, (size_t)group_cap * sizeof(GroupDesc)));
     HIP_TRY(hipHostMalloc(&s.h_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
     HIP_TRY(hipMalloc(&s.d_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
-    HIP_TRY(hipHostMalloc(&s.h_recn, (size_t)max_upd * 2));
     HIP_TRY(hipHostMalloc(&s.h_err, 4));
     HIP_TRY(hipHostMalloc(&s.h_descs, (size_t)max_upd * sizeof(UpdDesc)));
   }
@@ -590,7 +588,6 @@ GraEngine::~GraEngine() {
     if (s.h_groups) (void)hipHostFree(s.h_groups);
     if (s.h_rundescs) (void)hipHostFree(s.h_rundescs);
     if (s.d_rundescs) (void)hipFree(s.d_rundescs);
-    if (s.h_recn) (void)hipHostFree(s.h_recn);
     if (s.h_err) (void)hipHostFree(s.h_err);
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
@@ -767,6 +764,7 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
       std::lock_guard<std::mutex> lk(ss.mu);
       ss.poisoned = true;
       ss.next_seq = ss.durable_seq + 1;
+      ss.cnt_failures++; /* ≅ kReplicatorHandleResponseFailure */
     }
     for (int i = 0; i < kEventsPerTick; i++) put_event(t.ev[i]);
     sl.busy = false;
@@ -937,6 +935,14 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
      * the reference appends (rocksdb_wrapper.cpp:19-20) is WAL-only and
      * never reaches the memtable. Chained nodes re-serve with this ts. */
     if (e->opts.retain_log) retain_locked(e, ss, base, count, ts, rep, len);
+    ss.cnt_updates++;
+    ss.cnt_in_bytes += len; /* ≅ kReplicatorInBytes (replicated_db.cpp:409) */
+    if (ts != 0) { /* ≅ kReplicatorLatency (replicated_db.cpp:370-374) */
+      int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
+          std::chrono::system_clock::now().time_since_epoch()).count());
+      ss.lat_sum_ms += (uint64_t)(now_ms > ts ? now_ms - ts : 0);
+      ss.lat_n++;
+    }
   }
   std::lock_guard<std::mutex> lk(e->mu);
   if (e->stage_fill + len + 16 > e->opts.staging_bytes ||
@@ -1019,6 +1025,8 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
     out[*n_out].len = (uint32_t)ent.rep.size();
     off += ent.rep.size();
     (*n_out)++;
+    ss.cnt_served++;
+    ss.cnt_out_bytes += ent.rep.size();
   }
   if (*n_out > 0) { /* mode-1 ack: acked once sent (replicated_db.cpp:543-546) */
     uint64_t last = out[*n_out - 1].seq;
@@ -1029,6 +1037,20 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
     if (sent > ss.acked_sent) ss.acked_sent = sent;
   }
   ss.ack_cv.notify_all();
+  return GRA_OK;
+}
+
+int gra_db_counters(GraDb *db, GraDbCounters *out) {
+  ShardState &ss = db->e->shards[db->shard];
+  std::lock_guard<std::mutex> lk(ss.mu);
+  out->updates_applied = ss.cnt_updates;
+  out->in_bytes = ss.cnt_in_bytes;
+  out->apply_failures = ss.cnt_failures;
+  out->updates_served = ss.cnt_served;
+  out->out_bytes = ss.cnt_out_bytes;
+  out->latency_ms_sum = ss.lat_sum_ms;
+  out->latency_samples = ss.lat_n;
+  out->latest_seq = ss.poisoned ? ss.durable_seq : ss.next_seq - 1;
   return GRA_OK;
 }
 

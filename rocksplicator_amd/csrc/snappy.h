@@ -15,6 +15,9 @@
  */
 #pragma once
 #include <stdint.h>
+#ifdef __HIP__
+#include <hip/hip_runtime.h> /* uint4 for the cooperative device variant */
+#endif
 
 #include "wb_format.h" /* WB_HD */
 

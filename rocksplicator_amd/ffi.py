@@ -37,6 +37,12 @@ class GraEngineOpts(C.Structure):
     ]
 
 
+class GraDbCounters(C.Structure):
+    _fields_ = [(n, C.c_uint64) for n in (
+        "updates_applied", "in_bytes", "apply_failures", "updates_served",
+        "out_bytes", "latency_ms_sum", "latency_samples", "latest_seq")]
+
+
 class GraServedUpdate(C.Structure):
     _fields_ = [
         ("seq", C.c_uint64),
@@ -116,6 +122,7 @@ def load():
                                     C.POINTER(C.c_uint32), C.c_char_p,
                                     C.c_size_t]
     lib.gra_wait_ack.argtypes = [C.c_void_p, C.c_uint64, C.c_int, C.c_int]
+    lib.gra_db_counters.argtypes = [C.c_void_p, C.POINTER(GraDbCounters)]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
     lib.gra_pin_free.argtypes = [C.c_void_p, C.POINTER(C.c_uint8)]
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
@@ -321,6 +328,13 @@ class Db:
         if rc != GRA_OK:
             raise RuntimeError(f"gra_get rc={rc}: {last_error(self.lib)}")
         return buf.raw[: vlen.value]
+
+    def counters(self):
+        """Per-db stats ≅ the reference's per-db counter fan-out
+        (replicator_stats.cpp:33-102). Returns a dict of counter names."""
+        c = GraDbCounters()
+        self.lib.gra_db_counters(self.h, C.byref(c))
+        return {n: getattr(c, n) for n, _ in GraDbCounters._fields_}
 
     def wait_ack(self, seq, confirmed=True, timeout_ms=2000):
         """MaxNumberBox::wait equivalent: True when the downstream ack
