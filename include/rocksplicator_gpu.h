@@ -61,6 +61,10 @@ typedef struct GraEngineOpts {
   uint64_t log_bytes;      /* retained-log cap across shards (0 = 256 MiB);
                               oldest batches evicted first (WAL_ttl analog,
                               performance.cpp:99) */
+  int drain_host;          /* 1: eagerly drain each run's bytes to host
+                              memtable buffers at flush (the literal
+                              "drain to host memtables" mode; default 0 =
+                              device-resident store with lazy fetch) */
 } GraEngineOpts;
 
 void gra_engine_opts_init(GraEngineOpts *opts); /* fill defaults */

@@ -721,6 +721,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   return GRA_OK;
 }
 
+static int fetch_run_impl(GraEngine *e, Run &r);
+
 int GraEngine::ingest_one(TickRec &t, bool wait) {
   if (wait) {
     HIP_TRY(hipEventSynchronize(t.ev[7]));
@@ -796,7 +798,10 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     run->hdr_cur = rd.hdr_off;      /* absolute offsets into d_store */
     run->payload_cur = rd.payload_off;
     run->pay_rel_base = rd.pay_rel_base;
-    if (rd.n_entries > 0) ss.runs.push_back(std::move(run));
+    if (rd.n_entries > 0) {
+      if (opts.drain_host) (void)fetch_run_impl(this, *run); /* eager D2H */
+      ss.runs.push_back(std::move(run));
+    }
     ss.durable_seq = rd.last_seq > ss.durable_seq ? rd.last_seq : ss.durable_seq;
     stats.records += rd.n_entries;
     stats.payload_bytes += rd.payload_bytes;
@@ -1074,7 +1079,7 @@ int gra_flush(GraEngine *e) {
   return e->flush_locked();
 }
 
-static int fetch_run(GraEngine *e, Run &r) {
+static int fetch_run_impl(GraEngine *e, Run &r) {
   if (r.resident()) return GRA_OK;
   r.hdrs.resize((size_t)r.n_entries * sizeof(wb::RecHdr));
   r.payload.resize(r.payload_bytes);
@@ -1097,7 +1102,7 @@ int gra_get(GraDb *db, const void *key, size_t klen, void *buf, size_t cap,
     std::lock_guard<std::mutex> lk(ss.mu);
     runs = ss.runs;
     for (auto &r : runs) { /* lazy fetch under the shard lock */
-      int rc = fetch_run(e, *r);
+      int rc = fetch_run_impl(e, *r);
       if (rc != GRA_OK) return rc;
     }
   }
@@ -1119,6 +1124,8 @@ void gra_pin_free(GraEngine *e, uint8_t *ptr) {
   (void)e;
   (void)hipHostFree(ptr);
 }
+
+extern "C" void gra_replay_destroy(GraReplay *r);
 
 /* host_counts: batch record counts read from a HOST copy of the headers
  * (needed because with an external device arena the host cannot read the
@@ -1169,12 +1176,12 @@ int gra_upload(GraEngine *e, const uint8_t *arena, size_t arena_bytes,
       hipMemcpy(r->d_blobs, arena, arena_bytes, hipMemcpyHostToDevice) !=
           hipSuccess) {
     g_err = "gra_upload: blob allocation/H2D failed";
-    delete r;
+    gra_replay_destroy(r);
     return GRA_ERR;
   }
   int rc = upload_common(e, r, arena, arena_bytes, descs, n, nullptr);
   if (rc != GRA_OK) {
-    delete r;
+    gra_replay_destroy(r);
     return rc;
   }
   *out = r;
@@ -1195,7 +1202,7 @@ int gra_upload_dev(GraEngine *e, void *dev_arena, size_t arena_bytes,
   r->external_blobs = true;
   int rc = upload_common(e, r, nullptr, arena_bytes, descs, n, counts);
   if (rc != GRA_OK) {
-    delete r;
+    gra_replay_destroy(r);
     return rc;
   }
   *out = r;
@@ -1233,7 +1240,7 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
       hipMalloc(&r->d_comp, comp_bytes + 16) != hipSuccess ||
       hipMalloc(&r->d_snaptasks, n * sizeof(SnapTask)) != hipSuccess) {
     g_err = "gra_upload_snappy: allocation failed";
-    delete r;
+    gra_replay_destroy(r);
     return GRA_ERR;
   }
   if (hipMemcpy(r->d_comp, comp_arena, comp_bytes, hipMemcpyHostToDevice) !=
@@ -1241,12 +1248,12 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
       hipMemcpy(r->d_snaptasks, tasks.data(), n * sizeof(SnapTask),
                 hipMemcpyHostToDevice) != hipSuccess) {
     g_err = "gra_upload_snappy: H2D failed";
-    delete r;
+    gra_replay_destroy(r);
     return GRA_ERR;
   }
   int rc = upload_common(e, r, nullptr, scratch, udescs.data(), n, counts);
   if (rc != GRA_OK) {
-    delete r;
+    gra_replay_destroy(r);
     return rc;
   }
   *out = r;

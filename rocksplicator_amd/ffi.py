@@ -34,6 +34,7 @@ class GraEngineOpts(C.Structure):
         ("max_wb_records", C.c_uint32),
         ("retain_log", C.c_int),
         ("log_bytes", C.c_uint64),
+        ("drain_host", C.c_int),
     ]
 
 
@@ -220,7 +221,8 @@ class Engine:
     fallback exists for the follower apply path."""
 
     def __init__(self, nshards, device=-1, merge_op=0, store_ring=0,
-                 store_bytes=0, staging_bytes=0, retain_log=0, log_bytes=0):
+                 store_bytes=0, staging_bytes=0, retain_log=0, log_bytes=0,
+                 drain_host=0):
         self.lib = load()
         opts = GraEngineOpts()
         self.lib.gra_engine_opts_init(C.byref(opts))
@@ -229,6 +231,7 @@ class Engine:
         opts.merge_op = merge_op
         opts.store_ring = store_ring
         opts.retain_log = retain_log
+        opts.drain_host = drain_host
         if log_bytes:
             opts.log_bytes = log_bytes
         if store_bytes:
