@@ -452,7 +452,7 @@ struct GraEngine {
   uint8_t *h_stage[2] = {nullptr, nullptr};
   uint8_t *d_stage_blobs = nullptr;
   UpdDesc *d_stage_descs = nullptr;
-  hipEvent_t stage_free[2];
+  hipEvent_t stage_free[2] = {nullptr, nullptr};
   int stage_cur = 0;
   size_t stage_fill = 0;
   std::vector<GraUpdateDesc> stage_descs;
@@ -581,7 +581,7 @@ GraEngine::~GraEngine() {
   for (auto e : event_pool) (void)hipEventDestroy(e);
   for (int i = 0; i < 2; i++) {
     if (h_stage[i]) (void)hipHostFree(h_stage[i]);
-    (void)hipEventDestroy(stage_free[i]);
+    if (stage_free[i]) (void)hipEventDestroy(stage_free[i]);
   }
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
@@ -948,6 +948,14 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
       ss.lat_sum_ms += (uint64_t)(now_ms > ts ? now_ms - ts : 0);
       ss.lat_n++;
     }
+  }
+  if (len + 16 > e->opts.staging_bytes) {
+    /* a single Update larger than the staging buffer cannot be staged;
+     * refuse it (the reference's responses are bounded by max_updates x
+     * batch size — this is a misconfiguration, not a data error) */
+    std::lock_guard<std::mutex> lk2(ss.mu);
+    ss.next_seq = base; /* roll the optimistic assignment back */
+    return 0;
   }
   std::lock_guard<std::mutex> lk(e->mu);
   if (e->stage_fill + len + 16 > e->opts.staging_bytes ||
