@@ -411,6 +411,7 @@ struct TickRec {
   uint32_t n = 0, ngroups = 0;
   uint64_t blob_bytes = 0;
   bool h2d_timed = false;
+  uint32_t evmask = 0; /* which ev[i] were recorded this tick */
   hipEvent_t ev[kEventsPerTick];
 };
 
@@ -651,12 +652,20 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   t.blob_bytes = blob_bytes;
   t.h2d_timed = time_h2d;
   for (int i = 0; i < kEventsPerTick; i++) t.ev[i] = get_event();
+  static const bool detailed = [] {
+    const char *v = getenv("GRA_DETAILED_EVENTS");
+    return v && v[0] == '1';
+  }();
+  auto rec = [&](int i) {
+    t.evmask |= 1u << i;
+    return hipEventRecord(t.ev[i], stream);
+  };
 
   if (!d_groups_dev) {
     HIP_TRY(hipMemcpyAsync(d_groups, sl.h_groups, ngroups * sizeof(GroupDesc),
                            hipMemcpyHostToDevice, stream));
   }
-  HIP_TRY(hipEventRecord(t.ev[0], stream)); /* tick start */
+  HIP_TRY(rec(0)); /* tick start */
   if (h2d_src != nullptr) { /* PCIe-inclusive path: stage blobs (+descs) */
     HIP_TRY(hipMemcpyAsync(d_h2d_dst, h2d_src, h2d_bytes,
                            hipMemcpyHostToDevice, stream));
@@ -667,7 +676,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                              hipMemcpyHostToDevice, stream));
     }
   }
-  HIP_TRY(hipEventRecord(t.ev[1], stream)); /* after h2d */
+  if (h2d_src != nullptr) HIP_TRY(rec(1)); /* after h2d */
   uint32_t nb = (n + 255) / 256;
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena */
     uint32_t nb16 = (n * 16 + 255) / 256; /* 16 cooperative lanes per update */
@@ -675,23 +684,23 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                        d_snaptasks, d_descw, n, (uint8_t *)d_blobs, d_err_ring,
                        tick);
     HIP_TRY(hipGetLastError());
+    HIP_TRY(rec(8)); /* after snappy */
   }
-  HIP_TRY(hipEventRecord(t.ev[8], stream)); /* after snappy */
   hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, opts.max_wb_records, d_err_ring, tick,
                      d_partial, d_bsums, d_reccache);
   HIP_TRY(hipGetLastError());
-  HIP_TRY(hipEventRecord(t.ev[2], stream)); /* after decode(+scan1) */
+  if (detailed) HIP_TRY(rec(2)); /* after decode(+scan1) */
   hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, d_bsums, nb,
                      d_cursor, d_place, opts.store_bytes, opts.store_ring,
                      task_cap, d_err_ring, tick);
   HIP_TRY(hipGetLastError());
-  HIP_TRY(hipEventRecord(t.ev[3], stream)); /* after scan(+reserve) */
+  if (detailed) HIP_TRY(rec(3)); /* after scan(+reserve) */
   hipLaunchKernelGGL(k_emit, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, d_partial, d_bsums, d_place, d_store,
                      d_tasks, d_reccache);
   HIP_TRY(hipGetLastError());
-  HIP_TRY(hipEventRecord(t.ev[4], stream)); /* after emit */
+  HIP_TRY(rec(4)); /* after emit (pre-copy) */
   /* group width by average update size (micro_copy.hip: g32 wins >=512B) */
   if (blob_bytes / (n ? n : 1) >= 512) {
     hipLaunchKernelGGL((k_copy<32>), dim3(2048), dim3(256), 0, stream, d_blobs,
@@ -701,12 +710,12 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                        d_store, d_place, d_tasks);
   }
   HIP_TRY(hipGetLastError());
-  HIP_TRY(hipEventRecord(t.ev[5], stream)); /* after copy */
+  HIP_TRY(rec(5)); /* after copy */
   hipLaunchKernelGGL(k_rundesc, dim3((ngroups + 255) / 256), dim3(256), 0,
                      stream, groups_for_kernel, ngroups, d_descw, d_totals,
                      d_partial, d_bsums, n, nb, d_place, sl.d_rundescs);
   HIP_TRY(hipGetLastError());
-  HIP_TRY(hipEventRecord(t.ev[5 + 1], stream)); /* main-stream tick end */
+  HIP_TRY(rec(6)); /* main-stream tick end */
   /* publish run descriptors on the copyout stream, overlapped with the next
    * tick's kernels (per-slot device buffer: no hazard with slot reuse —
    * ingest waits on ev[7], which gates sl.busy) */
@@ -735,20 +744,37 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     }
   }
   Slot &sl = slots[t.slot];
-  /* stats */
+  /* stats: chained deltas over whichever events were recorded (fine-grained
+   * stage events are opt-in via GRA_DETAILED_EVENTS; coarse mode lumps
+   * decode+scan+emit into emit_ms) */
   float ms;
+  auto has = [&](int i) { return (t.evmask >> i) & 1u; };
   auto dt = [&](int a, int b) {
+    if (!has(a) || !has(b)) return 0.0;
     (void)hipEventElapsedTime(&ms, t.ev[a], t.ev[b]);
     return (double)ms;
   };
-  if (t.h2d_timed) stats.h2d_ms += dt(0, 1);
-  stats.snappy_ms += dt(1, 8);
-  stats.decode_ms += dt(8, 2);
-  stats.scan_ms += dt(2, 3);
-  stats.emit_ms += dt(3, 4);
+  int prev = 0;
+  if (has(1)) {
+    stats.h2d_ms += dt(0, 1);
+    prev = 1;
+  }
+  if (has(8)) {
+    stats.snappy_ms += dt(prev, 8);
+    prev = 8;
+  }
+  if (has(2)) {
+    stats.decode_ms += dt(prev, 2);
+    prev = 2;
+  }
+  if (has(3)) {
+    stats.scan_ms += dt(prev, 3);
+    prev = 3;
+  }
+  stats.emit_ms += dt(prev, 4);
   stats.copy_ms += dt(4, 5);
   stats.runfix_ms += dt(5, 6);
-  stats.total_ms += t.h2d_timed ? dt(0, 6) : dt(1, 6);
+  stats.total_ms += dt(0, 6);
   stats.ticks += 1;
   stats.updates += t.n;
   stats.blob_bytes += t.blob_bytes;
