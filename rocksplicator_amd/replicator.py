@@ -164,3 +164,27 @@ class Replicator:
         for rs in dbs:
             self._stop_pull(rs)
             rs.db.close()
+
+
+def dump_stats_text(replicator):
+    """Text stats dump ≅ common::Stats::DumpStatsAsText with the reference's
+    per-db counter naming scheme (replicator_stats.cpp:33-102 +
+    per-db suffixing as in replicator_stats.h:66-72)."""
+    lines = []
+    with replicator._mu:
+        items = list(replicator._dbs.items())
+    for name, rs in items:
+        c = rs.db.counters()
+        avg_lat = (c["latency_ms_sum"] / c["latency_samples"]
+                   if c["latency_samples"] else 0.0)
+        lines += [
+            f"replicator_in_updates_{name}: {c['updates_applied']}",
+            f"replicator_in_bytes_{name}: {c['in_bytes']}",
+            f"replicator_out_updates_{name}: {c['updates_served']}",
+            f"replicator_out_bytes_{name}: {c['out_bytes']}",
+            f"replicator_handle_response_failure_{name}: {c['apply_failures']}",
+            f"replicator_latency_ms_{name}: {avg_lat:.3f}",
+            f"latest_seq_{name}: {c['latest_seq']}",
+            f"role_{name}: {rs.role}",
+        ]
+    return "\n".join(lines)

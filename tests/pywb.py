@@ -75,5 +75,27 @@ class PyBatch:
         self.count += 1
         return self
 
+    def cf_delete(self, cf, k):
+        self.body += bytes([TYPE_CF_DELETION]) + varint32(cf) + lps(k)
+        self.count += 1
+        return self
+
+    def cf_merge(self, cf, k, v):
+        self.body += bytes([TYPE_CF_MERGE]) + varint32(cf) + lps(k) + lps(v)
+        self.count += 1
+        return self
+
+    def noop(self):
+        self.body += bytes([TYPE_NOOP])  # consumes no seq
+        return self
+
+    def begin_prepare(self):
+        self.body += bytes([0x09])  # marker, no payload, no seq
+        return self
+
+    def commit_xid(self, xid):
+        self.body += bytes([0x0B]) + lps(xid)  # xid slice, no seq
+        return self
+
     def data(self):
         return struct.pack("<QI", self.seq, self.count) + self.body

@@ -187,3 +187,16 @@ def test_write_degradation_counter():
     assert t1 - mid < 0.15   # 5 x ~10ms degraded
     rep.close()
     e.close()
+
+
+def test_stats_text_dump(olib):
+    from rocksplicator_amd.replicator import LEADER, Replicator, dump_stats_text
+    e = ra.Engine(nshards=2, retain_log=1)
+    rep = Replicator(e)
+    rep.add_db("shard_a", LEADER)
+    rep.write("shard_a", PyBatch().put(b"k", b"v").data())
+    txt = dump_stats_text(rep)
+    assert "latest_seq_shard_a: 1" in txt
+    assert "role_shard_a: LEADER" in txt
+    rep.close()
+    e.close()
