@@ -39,6 +39,7 @@
 #include <memory>
 #include <mutex>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "../../include/rocksplicator_gpu.h"
@@ -463,14 +464,24 @@ struct GraEngine {
   GroupDesc *d_groups = nullptr;
   uint32_t *d_err_ring = nullptr; /* kErrRing slots, zeroed at init */
   uint32_t tick_id = 0;
-  /* streaming-ingest staging (double-buffered pinned + device) */
-  uint8_t *h_stage[2] = {nullptr, nullptr};
+  /* streaming-ingest staging: double-buffered pinned, LOCK-FREE writer
+   * reservation (atomic byte/slot tickets; abandoned slots get len = 0 and
+   * are filtered at tick build) so concurrent HandleReplicateResponse
+   * callers never serialize on an engine lock */
+  struct StageBuf {
+    uint8_t *pin = nullptr;
+    GraUpdateDesc *descs = nullptr; /* plain host alloc, max_upd slots */
+    alignas(64) std::atomic<uint64_t> pos{0};
+    alignas(64) std::atomic<uint32_t> nslots{0};
+    alignas(64) std::atomic<int> writers{0};
+    alignas(64) std::atomic<bool> closed{false};
+    uint64_t epoch = 0; /* bumped at every swap: invalidates thread chunks */
+    hipEvent_t free_ev = nullptr; /* recorded after this buffer's H2D */
+  };
+  StageBuf stage[2];
+  std::atomic<StageBuf *> cur_stage{nullptr};
   uint8_t *d_stage_blobs = nullptr;
   UpdDesc *d_stage_descs = nullptr;
-  hipEvent_t stage_free[2] = {nullptr, nullptr};
-  int stage_cur = 0;
-  size_t stage_fill = 0;
-  std::vector<GraUpdateDesc> stage_descs;
   /* slots + pending ticks */
   Slot slots[kSlots];
   std::deque<TickRec> pending;
@@ -571,11 +582,18 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMemset(d_err_ring, 0, kErrRing * 4));
   HIP_TRY(hipMalloc(&d_stage_blobs, opts.staging_bytes + 16));
   HIP_TRY(hipMalloc(&d_stage_descs, (size_t)max_upd * sizeof(UpdDesc)));
+  /* epochs must be unique ACROSS engine instances: a thread-local staging
+   * chunk could otherwise match a freshly created engine reusing the same
+   * heap address (ABA) */
+  static std::atomic<uint64_t> g_epoch{1};
   for (int i = 0; i < 2; i++) {
-    HIP_TRY(hipHostMalloc(&h_stage[i], opts.staging_bytes + 16));
-    HIP_TRY(hipEventCreate(&stage_free[i]));
-    HIP_TRY(hipEventRecord(stage_free[i], stream));
+    HIP_TRY(hipHostMalloc(&stage[i].pin, opts.staging_bytes + 16));
+    stage[i].descs = (GraUpdateDesc *)malloc((size_t)max_upd * sizeof(GraUpdateDesc));
+    stage[i].epoch = g_epoch.fetch_add(1u << 20);
+    HIP_TRY(hipEventCreate(&stage[i].free_ev));
+    HIP_TRY(hipEventRecord(stage[i].free_ev, stream));
   }
+  cur_stage.store(&stage[0]);
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
     HIP_TRY(hipHostMalloc(&s.h_groups, (size_t)group_cap * sizeof(GroupDesc)));
@@ -595,8 +613,9 @@ GraEngine::~GraEngine() {
       if (t.ev[i]) (void)hipEventDestroy(t.ev[i]);
   for (auto e : event_pool) (void)hipEventDestroy(e);
   for (int i = 0; i < 2; i++) {
-    if (h_stage[i]) (void)hipHostFree(h_stage[i]);
-    if (stage_free[i]) (void)hipEventDestroy(stage_free[i]);
+    if (stage[i].pin) (void)hipHostFree(stage[i].pin);
+    free(stage[i].descs);
+    if (stage[i].free_ev) (void)hipEventDestroy(stage[i].free_ev);
   }
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
@@ -861,24 +880,44 @@ int GraEngine::ingest(bool wait_all) {
 }
 
 int GraEngine::stream_tick_locked() {
-  if (stage_descs.empty()) return GRA_OK;
-  uint32_t n = (uint32_t)stage_descs.size();
-  int cur = stage_cur;
-  /* counting-sort descs by shard (stable) and build UpdDesc + groups */
+  StageBuf *old = cur_stage.load(std::memory_order_acquire);
+  if (old->nslots.load(std::memory_order_relaxed) == 0) return GRA_OK;
+  StageBuf *next = old == &stage[0] ? &stage[1] : &stage[0];
+  /* the next buffer must have drained its previous H2D before writers
+   * reuse it */
+  HIP_TRY(hipEventSynchronize(next->free_ev));
+  next->pos.store(0, std::memory_order_relaxed);
+  next->nslots.store(0, std::memory_order_relaxed);
+  next->closed.store(false, std::memory_order_release);
+  /* swap: new writers land in `next`; then quiesce `old` */
+  old->closed.store(true, std::memory_order_release);
+  cur_stage.store(next, std::memory_order_release);
+  while (old->writers.load(std::memory_order_acquire) != 0)
+    std::this_thread::yield();
+  uint32_t nall = old->nslots.load(std::memory_order_relaxed);
+  if (nall > max_upd) nall = max_upd;
+  uint64_t fill = old->pos.load(std::memory_order_relaxed);
+  if (fill > opts.staging_bytes) fill = opts.staging_bytes;
+  /* counting-sort by shard, filtering abandoned (len==0) slots */
   std::vector<uint32_t> cnt(opts.nshards + 1, 0);
-  for (auto &d : stage_descs) cnt[d.shard + 1]++;
+  for (uint32_t i = 0; i < nall; i++)
+    if (old->descs[i].len) cnt[old->descs[i].shard + 1]++;
   for (uint32_t s = 0; s < opts.nshards; s++) cnt[s + 1] += cnt[s];
+  uint32_t n = cnt[opts.nshards];
+  if (n == 0) return GRA_OK;
   std::vector<UpdDesc> ud(n);
   std::vector<GroupDesc> groups;
   uint64_t blob_bytes = 0;
   {
     std::vector<uint32_t> pos = cnt;
-    for (auto &d : stage_descs) {
-      uint32_t i = pos[d.shard]++;
-      ud[i].off = d.off;
-      ud[i].len = d.len;
-      ud[i].shard = d.shard;
-      ud[i].base_seq = (uint64_t)d.ts; /* base_seq stashed at submit */
+    for (uint32_t i = 0; i < nall; i++) {
+      const GraUpdateDesc &d = old->descs[i];
+      if (!d.len) continue;
+      uint32_t j = pos[d.shard]++;
+      ud[j].off = d.off;
+      ud[j].len = d.len;
+      ud[j].shard = d.shard;
+      ud[j].base_seq = (uint64_t)d.ts; /* base_seq stashed at submit */
       blob_bytes += d.len;
     }
     for (uint32_t s = 0; s < opts.nshards; s++)
@@ -886,16 +925,10 @@ int GraEngine::stream_tick_locked() {
         groups.push_back({s, cnt[s], cnt[s + 1] - cnt[s], 0});
   }
   int rc = enqueue_tick(d_stage_blobs, d_stage_descs, n, groups, blob_bytes,
-                        true, h_stage[cur], stage_fill, d_stage_blobs,
-                        ud.data());
+                        true, old->pin, fill, d_stage_blobs, ud.data());
   if (rc != GRA_OK) return rc;
-  /* mark this pinned buffer reusable once the H2D completed */
-  HIP_TRY(hipEventRecord(stage_free[cur], stream));
-  stage_cur ^= 1;
-  stage_fill = 0;
-  stage_descs.clear();
-  /* wait until the other buffer's H2D has drained before reuse */
-  HIP_TRY(hipEventSynchronize(stage_free[stage_cur]));
+  /* mark the old pinned buffer reusable once its H2D completed */
+  HIP_TRY(hipEventRecord(old->free_ev, stream));
   return GRA_OK;
 }
 
@@ -996,22 +1029,68 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     ss.next_seq = base; /* roll the optimistic assignment back */
     return 0;
   }
-  std::lock_guard<std::mutex> lk(e->mu);
-  if (e->stage_fill + len + 16 > e->opts.staging_bytes ||
-      e->stage_descs.size() >= e->max_upd) {
-    int rc = e->stream_tick_locked();
-    if (rc != GRA_OK) return 0;
+  /* lock-free staging with per-thread CHUNK reservation: a thread grabs a
+   * ~256 KiB byte range + a run of desc slots in one pair of atomic adds
+   * and bump-allocates locally, so the shared cachelines are touched once
+   * per ~chunk instead of once per update. Reserved-but-unused desc slots
+   * are pre-zeroed (len = 0 -> filtered at tick build). */
+  struct Chunk {
+    GraEngine *e = nullptr;
+    GraEngine::StageBuf *sb = nullptr;
+    uint64_t epoch = 0;
+    uint64_t base = 0, used = 0, cap = 0;
+    uint32_t slot = 0, slots_left = 0;
+  };
+  thread_local Chunk ck;
+  constexpr uint64_t kChunkBytes = 256 << 10;
+  constexpr uint32_t kChunkSlots = 128;
+  for (int attempt = 0; attempt < 4000; attempt++) {
+    GraEngine::StageBuf *sb = e->cur_stage.load(std::memory_order_acquire);
+    sb->writers.fetch_add(1, std::memory_order_acq_rel);
+    if (sb->closed.load(std::memory_order_acquire)) {
+      sb->writers.fetch_sub(1, std::memory_order_acq_rel);
+      std::this_thread::yield();
+      continue;
+    }
+    bool chunk_ok = ck.e == e && ck.sb == sb && ck.epoch == sb->epoch &&
+                    ck.used + len <= ck.cap && ck.slots_left > 0;
+    if (!chunk_ok) {
+      uint64_t want = len > kChunkBytes ? (uint64_t)len : kChunkBytes;
+      uint64_t off = sb->pos.fetch_add(want, std::memory_order_relaxed);
+      uint32_t slot = sb->nslots.fetch_add(kChunkSlots, std::memory_order_relaxed);
+      if (off + want + 16 > e->opts.staging_bytes ||
+          slot + kChunkSlots > e->max_upd) {
+        /* buffer full: abandon (slots pre-zeroed below never happened —
+         * zero whatever part of the slot run is in range) */
+        for (uint32_t i = slot; i < slot + kChunkSlots && i < e->max_upd; i++)
+          sb->descs[i].len = 0;
+        ck.sb = nullptr;
+        sb->writers.fetch_sub(1, std::memory_order_acq_rel);
+        std::lock_guard<std::mutex> lk(e->mu);
+        if (e->cur_stage.load(std::memory_order_acquire) == sb) {
+          if (e->stream_tick_locked() != GRA_OK) return 0;
+        }
+        continue;
+      }
+      for (uint32_t i = 0; i < kChunkSlots; i++) sb->descs[slot + i].len = 0;
+      ck = {e, sb, sb->epoch, off, 0, want, slot, kChunkSlots};
+    }
+    uint64_t off = ck.base + ck.used;
+    memcpy(sb->pin + off, rep, len);
+    GraUpdateDesc d;
+    d.shard = db->shard;
+    d.len = (uint32_t)len;
+    d.off = off;
+    d.ts = (int64_t)base; /* staging path smuggles base_seq here */
+    sb->descs[ck.slot] = d;
+    ck.used += len;
+    ck.slot++;
+    ck.slots_left--;
+    sb->writers.fetch_sub(1, std::memory_order_release);
+    return 1;
   }
-  uint8_t *dst = e->h_stage[e->stage_cur] + e->stage_fill;
-  memcpy(dst, rep, len);
-  GraUpdateDesc d;
-  d.shard = db->shard;
-  d.len = (uint32_t)len;
-  d.off = e->stage_fill;
-  d.ts = (int64_t)base; /* staging path smuggles base_seq here */
-  e->stage_descs.push_back(d);
-  e->stage_fill += len;
-  return 1;
+  g_err = "staging contention: could not reserve a slot";
+  return 0;
 }
 
 uint64_t gra_latest_seq(GraDb *db) {
