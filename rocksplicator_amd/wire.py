@@ -1,0 +1,152 @@
+"""wire.py — TCP transport for the replication pull protocol.
+
+Restates the wire content of rocksdb_replicator/thrift/replicator.thrift
+(the ★ IDL row): ReplicateRequest{seq_no, db_name, max_wait_ms,
+max_updates, role} (:21-42) and ReplicateResponse{updates: [Update{raw_data,
+timestamp, seq_no}]} (:44-70), served by the leader's update read-out and
+consumed by the follower's pull loop. Framing is this framework's own
+length-prefixed binary (SURVEY allows "plain TCP or thrift-compatible
+framing" — fbthrift itself is RPC plumbing out of tier scope); the payload
+bytes on the wire are the exact rep blobs the engines exchange in-process.
+
+Long-poll semantics: the server holds a request up to max_wait_ms when it
+has no updates past seq_no (replicated_db.cpp long-poll via
+NonBlockingConditionVariable; here a threading.Condition signalled by a
+post-write hook).
+"""
+import socket
+import socketserver
+import struct
+import threading
+
+MAGIC = 0x47524150  # "GRAP"
+ROLE_FOLLOWER = 0
+ROLE_OBSERVER = 1
+
+_REQ = struct.Struct("<IQIIB")  # magic, seq_no, max_wait_ms, max_updates, role
+_UPD = struct.Struct("<QqI")    # seq, ts, len
+
+
+def _recv_exact(sock, n):
+    buf = b""
+    while len(buf) < n:
+        chunk = sock.recv(n - len(buf))
+        if not chunk:
+            raise ConnectionError("peer closed")
+        buf += chunk
+    return buf
+
+
+def encode_request(seq_no, db_name, max_wait_ms=10000, max_updates=50,
+                   role=ROLE_FOLLOWER):
+    name = db_name.encode()
+    return (_REQ.pack(MAGIC, seq_no, max_wait_ms, max_updates, role)
+            + struct.pack("<H", len(name)) + name)
+
+
+def decode_request(sock):
+    hdr = _recv_exact(sock, _REQ.size)
+    magic, seq_no, max_wait_ms, max_updates, role = _REQ.unpack(hdr)
+    if magic != MAGIC:
+        raise ValueError("bad magic")
+    (nlen,) = struct.unpack("<H", _recv_exact(sock, 2))
+    name = _recv_exact(sock, nlen).decode()
+    return seq_no, name, max_wait_ms, max_updates, role
+
+
+def encode_response(updates):
+    """updates: [(seq, ts, rep_bytes)] -> frame"""
+    out = [struct.pack("<I", len(updates))]
+    for seq, ts, rep in updates:
+        out.append(_UPD.pack(seq, ts, len(rep)))
+        out.append(rep)
+    return b"".join(out)
+
+
+def decode_response(sock):
+    (n,) = struct.unpack("<I", _recv_exact(sock, 4))
+    ups = []
+    for _ in range(n):
+        seq, ts, ln = _UPD.unpack(_recv_exact(sock, _UPD.size))
+        ups.append((seq, ts, _recv_exact(sock, ln)))
+    return ups
+
+
+class UpdateServer:
+    """Leader-side server ≅ ReplicatorHandler::async_tm_replicate
+    (replicator_handler.cpp:24-41) + handleReplicateRequest's long-poll."""
+
+    def __init__(self, host="127.0.0.1", port=0):
+        self._dbs = {}
+        self._cond = threading.Condition()
+        outer = self
+
+        class Handler(socketserver.BaseRequestHandler):
+            def handle(self):
+                try:
+                    while True:
+                        (seq_no, name, max_wait_ms, max_updates,
+                         _role) = decode_request(self.request)
+                        db = outer._dbs.get(name)
+                        if db is None:
+                            self.request.sendall(encode_response([]))
+                            continue
+                        ups = db.get_updates(seq_no, max_updates)
+                        if not ups and max_wait_ms:
+                            with outer._cond:  # long-poll
+                                outer._cond.wait(max_wait_ms / 1e3)
+                            ups = db.get_updates(seq_no, max_updates)
+                        self.request.sendall(encode_response(ups))
+                except (ConnectionError, OSError):
+                    pass
+
+        class Srv(socketserver.ThreadingTCPServer):
+            allow_reuse_address = True
+            daemon_threads = True
+
+        self._srv = Srv((host, port), Handler)
+        self.port = self._srv.server_address[1]
+        self._thread = threading.Thread(target=self._srv.serve_forever,
+                                        daemon=True)
+        self._thread.start()
+
+    def register(self, db_name, db):
+        self._dbs[db_name] = db
+
+    def notify_write(self):
+        """Wake long-polling pullers (≅ cond_var_.notifyAll after a leader
+        write, replicated_db.cpp:138)."""
+        with self._cond:
+            self._cond.notify_all()
+
+    def close(self):
+        self._srv.shutdown()
+        self._srv.server_close()
+
+
+class RemoteUpstream:
+    """Client-side handle usable wherever an in-process upstream db is:
+    exposes get_updates(since, max) over the wire, so replicator.pull_once /
+    Replicator pull threads work unchanged against a TCP leader."""
+
+    def __init__(self, host, port, db_name, max_wait_ms=0):
+        self._sock = socket.create_connection((host, port))
+        self._sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        self._name = db_name
+        self._max_wait_ms = max_wait_ms
+        self._mu = threading.Lock()
+
+    def get_updates(self, since_seq, max_updates=50):
+        with self._mu:
+            self._sock.sendall(encode_request(since_seq, self._name,
+                                              self._max_wait_ms, max_updates))
+            return decode_response(self._sock)
+
+    def latest_seq(self):
+        """Probe: ask for nothing past a huge seq is wrong; instead request
+        from 0 with max 0? The pull loop only needs get_updates; for catch-up
+        checks, peek one update past `since`."""
+        raise NotImplementedError("pull loops use get_updates only")
+
+    def close(self):
+        self._sock.close()
