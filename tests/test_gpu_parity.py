@@ -484,3 +484,26 @@ def test_fuzz_all_record_types(olib):
                 ck = cf.to_bytes(4, "little") + k
                 assert dbs[s].get(ck) == ost.get(s, ck), (s, cf, k)
     e.close()
+
+
+@pytest.mark.parametrize("klen,vlen", [
+    (7, 100),    # tiny unaligned keys; odd values
+    (24, 1000),  # value dst unaligned (24 % 16 != 0) -> dword-funnel fallback
+    (40, 512),   # keys > 32B -> key copy-task path (not inlined anywhere)
+    (3, 17),     # everything misaligned and tiny
+])
+def test_replay_parity_odd_shapes(olib, klen, vlen):
+    nupd, nshards = 15000, 32
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=klen, val_len=vlen, kind=2,
+                                       key_space=1 << 14, seed=400 + klen)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd)
+    e = ra.Engine(nshards=nshards)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    for first in range(0, nupd, 5000):
+        rep.tick(first, 5000)
+    rep.sync()
+    keys = collect_keys(olib, raw, descs, 4000, per_shard_cap=24)
+    check_parity(e, ost, keys, range(0, nshards, 3))
+    e.close()

@@ -312,3 +312,40 @@ def test_wire_format_roundtrip():
     ups = [(1, 99, b"\x00" * 12), (2, 100, b"payload-bytes")]
     frame = wire.encode_response(ups)
     assert wire.decode_response(FakeSock(frame)) == ups
+
+
+def test_concurrent_write_and_pull_soak(olib):
+    """Writers and a puller race on the same shard: leader writes stream in
+    while the follower pulls continuously (log eviction, ack box and
+    serving all exercised mid-mutation). Ends with exact convergence."""
+    import threading
+    leader = ra.Engine(nshards=2, merge_op=1, retain_log=1, log_bytes=64 << 20)
+    follower = ra.Engine(nshards=2, merge_op=1)
+    ldb, fdb = leader.open(0), follower.open(0)
+    one = (1).to_bytes(8, "little")
+    N = 20000
+    stop = threading.Event()
+
+    def writer():
+        for i in range(N):
+            ldb.write_leader(PyBatch().merge(b"ctr", one).data())
+        stop.set()
+
+    def puller():
+        while not (stop.is_set() and fdb.latest_seq() >= N):
+            if not replicator.pull_once(ldb, fdb):
+                follower.flush()
+        follower.flush()
+
+    tw = threading.Thread(target=writer)
+    tp = threading.Thread(target=puller)
+    tw.start()
+    tp.start()
+    tw.join(timeout=120)
+    tp.join(timeout=120)
+    assert not tw.is_alive() and not tp.is_alive()
+    assert fdb.latest_seq() == ldb.latest_seq() == N
+    assert int.from_bytes(fdb.get(b"ctr"), "little") == N
+    assert int.from_bytes(ldb.get(b"ctr"), "little") == N
+    leader.close()
+    follower.close()
