@@ -294,26 +294,6 @@ def test_tcp_wire_replication(olib):
     follower.close()
 
 
-def test_wire_format_roundtrip():
-    """Framing known-answer: request/response encode/decode symmetry."""
-    from rocksplicator_amd import wire
-    import io
-
-    class FakeSock:
-        def __init__(self, data):
-            self.b = io.BytesIO(data)
-
-        def recv(self, n):
-            return self.b.read(n)
-
-    req = wire.encode_request(12345, "shard_07", 10000, 50, wire.ROLE_FOLLOWER)
-    seq, name, wait, maxu, role = wire.decode_request(FakeSock(req))
-    assert (seq, name, wait, maxu, role) == (12345, "shard_07", 10000, 50, 0)
-    ups = [(1, 99, b"\x00" * 12), (2, 100, b"payload-bytes")]
-    frame = wire.encode_response(ups)
-    assert wire.decode_response(FakeSock(frame)) == ups
-
-
 def test_concurrent_write_and_pull_soak(olib):
     """Writers and a puller race on the same shard: leader writes stream in
     while the follower pulls continuously (log eviction, ack box and
