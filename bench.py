@@ -84,11 +84,10 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s, snappy=False)
     bench_fn = (olib.orc_cpu_snappy_apply_bench if snappy
                 else olib.orc_cpu_apply_bench)
 
-    def run(sample_n, threads):
+    def run(sample_n, threads, fn):
         st = olib.orc_store_create(nshards, 0)
         try:
-            secs = bench_fn(
-                st, C.cast(raw_arena, C.c_void_p), ods, sample_n, threads)
+            secs = fn(st, C.cast(raw_arena, C.c_void_p), ods, sample_n, threads)
         finally:
             olib.orc_store_destroy(st)
         return secs
@@ -97,12 +96,12 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s, snappy=False)
     total_n, total_s, reps = 0, 0.0, 0
     sample_n = min(n, 2_000_000)
     while total_s < target_s and reps < 64:
-        secs = run(sample_n, ncores)
+        secs = run(sample_n, ncores, bench_fn)
         total_n += sample_n
         total_s += secs
         reps += 1
     rate = total_n / total_s if total_s > 0 else 0.0
-    return {
+    out = {
         "value": rate,
         "unit": "updates/s",
         "cores": ncores,
@@ -111,6 +110,12 @@ def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s, snappy=False)
                   + (" (snappy decompress+apply)" if snappy else "") +
                   f", {ncores} threads, {total_s:.2f}s total",
     }
+    if not snappy:
+        # WAL-on leg (both variants reported); `value` stays the
+        # faster WAL-less leg — conservative for the GPU/CPU ratio
+        wal_secs = run(sample_n, ncores, olib.orc_cpu_apply_bench_wal)
+        out["wal_on_value"] = sample_n / wal_secs if wal_secs > 0 else 0.0
+    return out
 
 
 def run_repartition(args, rank, world, local, dist):

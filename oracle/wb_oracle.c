@@ -531,22 +531,40 @@ typedef struct {
   const uint8_t *arena;
   const OrcUpdateDesc *descs;
   uint64_t n;
-  int tid, nthreads;
+  int tid, nthreads, wal;
 } BenchArg;
 
 static void *bench_worker(void *p) {
   BenchArg *a = (BenchArg *)p;
+  /* WAL-on variant: the reference's DB::Write appends the rep (+LogData
+   * trailer) to the WAL before the memtable insert (rocksdb_wrapper.cpp:22);
+   * modeled as a per-thread log append (no fsync — group-commit steady
+   * state), reported alongside the WAL-less variant per SURVEY §8d. */
+  uint8_t *wal_buf = NULL;
+  size_t wal_len = 0, wal_cap = 0;
   /* per-shard sequential, cross-shard concurrent: thread owns shard % nthreads */
   for (uint64_t i = 0; i < a->n; i++) {
     const OrcUpdateDesc *d = &a->descs[i];
     if ((int)(d->shard % (uint32_t)a->nthreads) != a->tid) continue;
+    if (a->wal) {
+      if (wal_len + d->len + 10 > wal_cap) {
+        wal_cap = wal_cap ? wal_cap * 2 : (1 << 20);
+        if (wal_cap < wal_len + d->len + 10) wal_cap = wal_len + d->len + 10;
+        wal_buf = (uint8_t *)realloc(wal_buf, wal_cap);
+      }
+      memcpy(wal_buf + wal_len, a->arena + d->off, d->len);
+      wal_len += d->len + 10; /* + LogData(ts) trailer bytes */
+      if (wal_len > (64u << 20)) wal_len = 0; /* rotated segment */
+    }
     orc_apply(a->s, d->shard, a->arena + d->off, d->len, d->ts);
   }
+  free(wal_buf);
   return NULL;
 }
 
-double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena, const OrcUpdateDesc *descs,
-                           uint64_t n, int nthreads) {
+static double cpu_bench(OrcStore *s, const uint8_t *arena,
+                        const OrcUpdateDesc *descs, uint64_t n, int nthreads,
+                        int wal) {
   if (nthreads < 1) nthreads = 1;
   pthread_t th[256];
   BenchArg args[256];
@@ -554,12 +572,23 @@ double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena, const OrcUpdateDes
   struct timespec t0, t1;
   clock_gettime(CLOCK_MONOTONIC, &t0);
   for (int i = 0; i < nthreads; i++) {
-    args[i] = (BenchArg){s, arena, descs, n, i, nthreads};
+    args[i] = (BenchArg){s, arena, descs, n, i, nthreads, wal};
     pthread_create(&th[i], NULL, bench_worker, &args[i]);
   }
   for (int i = 0; i < nthreads; i++) pthread_join(th[i], NULL);
   clock_gettime(CLOCK_MONOTONIC, &t1);
   return (t1.tv_sec - t0.tv_sec) + (t1.tv_nsec - t0.tv_nsec) * 1e-9;
+}
+
+double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena,
+                           const OrcUpdateDesc *descs, uint64_t n, int nthreads) {
+  return cpu_bench(s, arena, descs, n, nthreads, 0);
+}
+
+double orc_cpu_apply_bench_wal(OrcStore *s, const uint8_t *arena,
+                               const OrcUpdateDesc *descs, uint64_t n,
+                               int nthreads) {
+  return cpu_bench(s, arena, descs, n, nthreads, 1);
 }
 
 /* ================= Snappy block format ================= */

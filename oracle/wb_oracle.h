@@ -120,6 +120,11 @@ typedef struct {
 } OrcUpdateDesc;
 double orc_cpu_apply_bench(OrcStore *s, const uint8_t *arena,
                            const OrcUpdateDesc *descs, uint64_t n, int nthreads);
+/* WAL-on variant (rep + LogData trailer appended to a rotated log buffer
+ * before the memtable insert — SURVEY §8d asks for both legs). */
+double orc_cpu_apply_bench_wal(OrcStore *s, const uint8_t *arena,
+                               const OrcUpdateDesc *descs, uint64_t n,
+                               int nthreads);
 
 
 /* ---------- Snappy block format (transport compression, config #5) ----------

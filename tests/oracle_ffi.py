@@ -67,6 +67,10 @@ def load():
     lib.orc_cpu_apply_bench.argtypes = [
         C.c_void_p, C.c_void_p, C.POINTER(OrcUpdateDesc), C.c_uint64, C.c_int,
     ]
+    lib.orc_cpu_apply_bench_wal.restype = C.c_double
+    lib.orc_cpu_apply_bench_wal.argtypes = [
+        C.c_void_p, C.c_void_p, C.POINTER(OrcUpdateDesc), C.c_uint64, C.c_int,
+    ]
     lib.orc_cpu_snappy_apply_bench.restype = C.c_double
     lib.orc_cpu_snappy_apply_bench.argtypes = [
         C.c_void_p, C.c_void_p, C.POINTER(OrcUpdateDesc), C.c_uint64, C.c_int,
