@@ -156,7 +156,8 @@ __global__ void k_decode(const uint8_t *__restrict__ blobs,
                          wb::WalkTotals *__restrict__ totals, uint32_t max_rec,
                          uint32_t *__restrict__ err_ring, uint32_t tick,
                          uint2 *__restrict__ partial, uint2 *__restrict__ bsums,
-                         wb::Rec *__restrict__ reccache) {
+                         wb::Rec *__restrict__ reccache,
+                         uint8_t *__restrict__ ok_out) {
   __shared__ uint2 sh[256];
   uint32_t i = blockIdx.x * 256 + threadIdx.x;
   uint2 v = make_uint2(0, 0);
@@ -169,6 +170,7 @@ __global__ void k_decode(const uint8_t *__restrict__ blobs,
                                   });
     if (t.n_records > max_rec) t.ok = 0;
     totals[i] = t;
+    ok_out[i] = (uint8_t)t.ok;
     if (!t.ok) atomicAdd(&err_ring[tick % kErrRing], 1u);
     if (t.ok) v = make_uint2(t.n_records, t.payload16);
   }
@@ -427,6 +429,7 @@ struct TickRec {
   uint64_t blob_bytes = 0;
   bool h2d_timed = false;
   uint32_t evmask = 0; /* which ev[i] were recorded this tick */
+  std::vector<uint32_t> counts; /* per-update record counts (error recovery) */
   hipEvent_t ev[kEventsPerTick];
 };
 
@@ -435,6 +438,8 @@ struct Slot {
   DevRunDesc *h_rundescs = nullptr;
   DevRunDesc *d_rundescs = nullptr; /* per-slot device buffer so the D2H can
                                        overlap the next tick's kernels */
+  uint8_t *d_ok = nullptr;         /* per-update validity from k_decode */
+  uint8_t *h_ok = nullptr;         /* pinned mirror (read only on error) */
   uint32_t *h_err = nullptr;
   UpdDesc *h_descs = nullptr;      /* staging-path desc upload */
   bool busy = false;
@@ -500,7 +505,8 @@ struct GraEngine {
                    const UpdDesc *h_descs_src = nullptr,
                    const GroupDesc *d_groups_dev = nullptr,
                    const uint8_t *d_comp = nullptr,
-                   const SnapTask *d_snaptasks = nullptr);
+                   const SnapTask *d_snaptasks = nullptr,
+                   std::vector<uint32_t> &&counts = {});
   int ingest(bool wait_all);
   int ingest_one(TickRec &t, bool wait);
   int flush_locked();
@@ -599,6 +605,8 @@ int GraEngine::init(const GraEngineOpts &o) {
     HIP_TRY(hipHostMalloc(&s.h_groups, (size_t)group_cap * sizeof(GroupDesc)));
     HIP_TRY(hipHostMalloc(&s.h_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
     HIP_TRY(hipMalloc(&s.d_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
+    HIP_TRY(hipMalloc(&s.d_ok, (size_t)max_upd));
+    HIP_TRY(hipHostMalloc(&s.h_ok, (size_t)max_upd));
     HIP_TRY(hipHostMalloc(&s.h_err, 4));
     HIP_TRY(hipHostMalloc(&s.h_descs, (size_t)max_upd * sizeof(UpdDesc)));
   }
@@ -622,6 +630,8 @@ GraEngine::~GraEngine() {
     if (s.h_groups) (void)hipHostFree(s.h_groups);
     if (s.h_rundescs) (void)hipHostFree(s.h_rundescs);
     if (s.d_rundescs) (void)hipFree(s.d_rundescs);
+    if (s.d_ok) (void)hipFree(s.d_ok);
+    if (s.h_ok) (void)hipHostFree(s.h_ok);
     if (s.h_err) (void)hipHostFree(s.h_err);
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
@@ -656,7 +666,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                             uint8_t *d_h2d_dst, const UpdDesc *h_descs_src,
                             const GroupDesc *d_groups_dev,
                             const uint8_t *d_comp,
-                            const SnapTask *d_snaptasks) {
+                            const SnapTask *d_snaptasks,
+                            std::vector<uint32_t> &&counts) {
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
@@ -684,6 +695,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   t.ngroups = ngroups;
   t.blob_bytes = blob_bytes;
   t.h2d_timed = time_h2d;
+  t.counts = std::move(counts);
   for (int i = 0; i < kEventsPerTick; i++) t.ev[i] = get_event();
   static const bool detailed = [] {
     const char *v = getenv("GRA_DETAILED_EVENTS");
@@ -720,7 +732,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   }
   hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, opts.max_wb_records, d_err_ring, tick,
-                     d_partial, d_bsums, d_reccache);
+                     d_partial, d_bsums, d_reccache, sl.d_ok);
   HIP_TRY(hipGetLastError());
   if (detailed) HIP_TRY(rec(2)); /* after decode(+scan1) */
   hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, d_bsums, nb,
@@ -757,6 +769,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                          hipMemcpyDeviceToHost, copyout));
   HIP_TRY(hipMemcpyAsync(sl.h_err, d_err_ring + (tick % kErrRing), 4,
                          hipMemcpyDeviceToHost, copyout));
+  HIP_TRY(hipMemcpyAsync(sl.h_ok, sl.d_ok, n, hipMemcpyDeviceToHost, copyout));
   HIP_TRY(hipEventRecord(t.ev[7], copyout)); /* publication done */
   pending.push_back(std::move(t));
   return GRA_OK;
@@ -812,16 +825,65 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
   stats.blob_bytes += t.blob_bytes;
   uint32_t err = *sl.h_err;
   if (err != 0) {
-    /* Rare corruption path: poison affected shards. Without per-update recn
-     * recovery (TODO round 2: precise truncation), poison every shard in the
-     * tick whose run can't be trusted. The tick's runs are still linked for
-     * the records that were emitted (failed updates emit nothing), but seqs
-     * after a failed update in the same shard are optimistic — roll those
-     * shards back hard: drop this tick's runs for them. */
+    /* Precise corruption recovery: per-update validity came back with the
+     * run descriptors (sl.h_ok). Per shard, keep the prefix of records from
+     * updates BEFORE its first corrupt one (their seqs are final), drop the
+     * rest (their host-assigned seqs assumed the corrupt batch applied),
+     * roll the shard's seq back to the kept boundary and poison it — the
+     * reference's failed-apply -> re-pull-from-LatestSequenceNumber cadence
+     * (replicated_db.cpp:378-382). Groups are in tick order, so a later
+     * group of an already-poisoned shard is dropped whole. */
+    const bool have_counts = t.counts.size() == t.n;
     for (uint32_t g = 0; g < t.ngroups; g++) {
       DevRunDesc &rd = sl.h_rundescs[g];
+      const GroupDesc &gd = sl.h_groups[g];
       ShardState &ss = shards[rd.shard];
       std::lock_guard<std::mutex> lk(ss.mu);
+      uint32_t keep_recs = 0;
+      uint64_t keep_seq = rd.base_seq - 1;
+      bool bad = ss.poisoned; /* shard already failed earlier this tick */
+      if (!bad) {
+        for (uint32_t i = 0; i < gd.n_upds; i++) {
+          if (!sl.h_ok[gd.first + i]) {
+            bad = true;
+            break;
+          }
+          uint32_t c = have_counts ? t.counts[gd.first + i] : 0;
+          keep_recs += c;
+          keep_seq += c;
+        }
+      }
+      if (!bad) { /* whole group clean */
+        auto run = std::make_shared<Run>();
+        run->base_seq = rd.base_seq;
+        run->last_seq = rd.last_seq;
+        run->n_entries = rd.n_entries;
+        run->payload_bytes = rd.payload_bytes;
+        run->hdr_cur = rd.hdr_off;
+        run->payload_cur = rd.payload_off;
+        run->pay_rel_base = rd.pay_rel_base;
+        if (opts.drain_host) (void)fetch_run_impl(this, *run);
+        if (rd.n_entries > 0 && !opts.store_ring) ss.runs.push_back(std::move(run));
+        if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
+        stats.records += rd.n_entries;
+        stats.payload_bytes += rd.payload_bytes;
+        continue;
+      }
+      if (keep_recs > 0 && have_counts && !opts.store_ring) {
+        /* truncated run: the leading clean records are contiguous */
+        auto run = std::make_shared<Run>();
+        run->base_seq = rd.base_seq;
+        run->last_seq = keep_seq;
+        run->n_entries = keep_recs;
+        run->payload_bytes = rd.payload_bytes; /* upper bound; fetch uses hdrs */
+        run->hdr_cur = rd.hdr_off;
+        run->payload_cur = rd.payload_off;
+        run->pay_rel_base = rd.pay_rel_base;
+        if (opts.drain_host) (void)fetch_run_impl(this, *run);
+        ss.runs.push_back(std::move(run));
+        stats.records += keep_recs;
+      }
+      if (keep_seq > ss.durable_seq) ss.durable_seq = keep_seq;
       ss.poisoned = true;
       ss.next_seq = ss.durable_seq + 1;
       ss.cnt_failures++; /* ≅ kReplicatorHandleResponseFailure */
@@ -924,8 +986,14 @@ int GraEngine::stream_tick_locked() {
       if (cnt[s + 1] > cnt[s])
         groups.push_back({s, cnt[s], cnt[s + 1] - cnt[s], 0});
   }
+  /* per-update counts from consecutive base seqs are unavailable here;
+   * recover them from the staged batch headers (4B read per update) */
+  std::vector<uint32_t> counts(n);
+  for (uint32_t i = 0; i < n; i++)
+    counts[i] = wb::fixed32_le(old->pin + ud[i].off + 8);
   int rc = enqueue_tick(d_stage_blobs, d_stage_descs, n, groups, blob_bytes,
-                        true, old->pin, fill, d_stage_blobs, ud.data());
+                        true, old->pin, fill, d_stage_blobs, ud.data(), nullptr,
+                        nullptr, nullptr, std::move(counts));
   if (rc != GRA_OK) return rc;
   /* mark the old pinned buffer reusable once its H2D completed */
   HIP_TRY(hipEventRecord(old->free_ev, stream));
@@ -1437,11 +1505,14 @@ int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
     return GRA_ERR;
   }
   TickPlan &plan = plan_for(r, first, n);
+  std::vector<uint32_t> counts(r->counts.begin() + first,
+                               r->counts.begin() + first + n);
   std::lock_guard<std::mutex> lk(e->mu);
   return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n,
                          plan.groups, plan.blob_bytes, false, nullptr, 0,
                          nullptr, nullptr, plan.d_groups, r->d_comp,
-                         r->snappy ? r->d_snaptasks + first : nullptr);
+                         r->snappy ? r->d_snaptasks + first : nullptr,
+                         std::move(counts));
 }
 
 int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {

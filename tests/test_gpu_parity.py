@@ -507,3 +507,43 @@ def test_replay_parity_odd_shapes(olib, klen, vlen):
     keys = collect_keys(olib, raw, descs, 4000, per_shard_cap=24)
     check_parity(e, ost, keys, range(0, nshards, 3))
     e.close()
+
+
+def test_precise_corruption_truncation(olib):
+    """A corrupt batch mid-tick must affect ONLY its shard, and that shard
+    keeps exactly the records from batches before the corrupt one."""
+    e = ra.Engine(nshards=3)
+    dbs = [e.open(s) for s in range(3)]
+    ost = oracle_ffi.Store(olib, 3)
+    # shard 0: clean stream; shard 1: good, good, BAD, good; shard 2: clean
+    for i in range(5):
+        for s in (0, 2):
+            rep = PyBatch().put(f"s{s}k{i}".encode(), b"v").data()
+            assert dbs[s].handle_replicate_response(rep)
+            assert ost.apply(s, rep)
+    good1 = [PyBatch().put(f"g{i}".encode(), f"w{i}".encode()).data()
+             for i in range(4)]
+    bad = bytearray(PyBatch().put(b"bad", b"bad").data())
+    bad[8] = 7  # count mismatch
+    assert dbs[1].handle_replicate_response(good1[0])
+    assert dbs[1].handle_replicate_response(good1[1])
+    assert dbs[1].handle_replicate_response(bytes(bad))
+    assert dbs[1].handle_replicate_response(good1[2])  # post-bad: dropped
+    e.flush()
+    # clean shards unaffected and parity-green
+    for s in (0, 2):
+        assert dbs[s].latest_seq() == ost.latest_seq(s) == 5
+        for i in range(5):
+            assert dbs[s].get(f"s{s}k{i}".encode()) == b"v"
+    # shard 1: exactly the 2 pre-corruption batches kept, then poisoned
+    assert dbs[1].latest_seq() == 2
+    assert dbs[1].get(b"g0") == b"w0" and dbs[1].get(b"g1") == b"w1"
+    assert dbs[1].get(b"g2") is None  # post-bad batch rolled back
+    assert not dbs[1].handle_replicate_response(good1[2])  # fail-once signal
+    # re-pull from the durable boundary succeeds
+    assert dbs[1].handle_replicate_response(good1[2])
+    assert dbs[1].handle_replicate_response(good1[3])
+    e.flush()
+    assert dbs[1].latest_seq() == 4
+    assert dbs[1].get(b"g2") == b"w2" and dbs[1].get(b"g3") == b"w3"
+    e.close()
