@@ -372,13 +372,18 @@ int orc_apply(OrcStore *s, uint32_t shard, const uint8_t *rep, size_t len, int64
   if (shard >= s->nshards) return 0;
   ShardTable *t = &s->shards[shard];
   /* two-pass: validate fully, then apply (a corrupt batch applies nothing —
-   * DB::Write fails before memtable insert on iterate corruption). */
+   * DB::Write fails before memtable insert on iterate corruption).
+   * Stack storage for typical batches: per-call malloc would handicap the
+   * CPU-baseline leg (the reference's memtable insert allocates from an
+   * arena, not malloc-per-op). */
   uint32_t nrec = 0, cnt = 0;
   uint64_t hdr_seq = 0;
   if (orc_decode(rep, len, NULL, 0, &nrec, &hdr_seq, &cnt) != 0) return 0;
-  OrcRecord *recs = (OrcRecord *)malloc(nrec * sizeof(OrcRecord));
+  OrcRecord stack_recs[64];
+  OrcRecord *recs = nrec <= 64 ? stack_recs
+                               : (OrcRecord *)malloc(nrec * sizeof(OrcRecord));
   if (orc_decode(rep, len, recs, nrec, &nrec, &hdr_seq, &cnt) != 0) {
-    free(recs);
+    if (recs != stack_recs) free(recs);
     return 0;
   }
   /* Follower assigns its own seqs: base = latest+1
@@ -409,7 +414,10 @@ int orc_apply(OrcStore *s, uint32_t shard, const uint8_t *rep, size_t len, int64
        * no column families; synthetic streams use cf 0). */
       if (r->cf_id != 0) {
         uint8_t tmp[4 + 65536];
-        if (r->key_len > 65536) { free(recs); return 0; }
+        if (r->key_len > 65536) {
+          if (recs != stack_recs) free(recs);
+          return 0;
+        }
         memcpy(tmp, &r->cf_id, 4);
         memcpy(tmp + 4, rep + r->key_off, r->key_len);
         shard_insert(t, bt, seq, tmp, 4 + r->key_len, rep + r->val_off, r->val_len);
@@ -418,7 +426,7 @@ int orc_apply(OrcStore *s, uint32_t shard, const uint8_t *rep, size_t len, int64
       }
     }
   }
-  free(recs);
+  if (recs != stack_recs) free(recs);
   t->latest_seq += cnt; /* Write consumes Count() seqs (assumption test :179-187) */
   return 1;
 }
