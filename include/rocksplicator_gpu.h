@@ -128,9 +128,12 @@ typedef struct {
   int64_t ts;     /* timestamp the leader stamped (ms) */
   uint32_t off, len;
 } GraServedUpdate;
+/* requester_role: 0 = FOLLOWER (the request's seq_no is posted as the
+ * confirmed ack), 1 = OBSERVER (no ack — replicated_db.cpp:452-456 ignores
+ * observer sequence numbers). */
 int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
                     GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
-                    size_t cap);
+                    size_t cap, int requester_role);
 
 /* Per-db counters ≅ the replicator stats hooks' per-db fan-out
  * (rocksdb_replicator/replicator_stats.cpp:33-102: replicator_in_bytes,

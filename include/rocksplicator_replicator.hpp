@@ -39,14 +39,15 @@ struct WireUpdate {
 using UpstreamFn =
     std::function<std::vector<WireUpdate>(uint64_t since, uint32_t max_updates)>;
 
-/* Serve an in-process leader GraDb as an upstream (test/demo wiring). */
-inline UpstreamFn local_upstream(GraDb *db) {
-  return [db](uint64_t since, uint32_t max_updates) {
+/* Serve an in-process leader GraDb as an upstream (test/demo wiring).
+ * observer: the request carries the observer role (no ACK posted). */
+inline UpstreamFn local_upstream(GraDb *db, bool observer = false) {
+  return [db, observer](uint64_t since, uint32_t max_updates) {
     std::vector<GraServedUpdate> out(max_updates);
     std::vector<uint8_t> buf(4u << 20);
     uint32_t n = 0;
     if (gra_get_updates(db, since, max_updates, out.data(), &n, buf.data(),
-                        buf.size()) != GRA_OK)
+                        buf.size(), observer ? 1 : 0) != GRA_OK)
       throw std::runtime_error(gra_last_error());
     std::vector<WireUpdate> ups(n);
     for (uint32_t i = 0; i < n; i++) {

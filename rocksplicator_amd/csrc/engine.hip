@@ -1195,7 +1195,7 @@ int gra_write_leader(GraDb *db, const uint8_t *rep, size_t len,
 
 int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
                     GraServedUpdate *out, uint32_t *n_out, uint8_t *buf,
-                    size_t cap) {
+                    size_t cap, int requester_role) {
   GraEngine *e = db->e;
   if (!e->opts.retain_log) {
     g_err = "retain_log disabled on this engine";
@@ -1204,8 +1204,10 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
   ShardState &ss = e->shards[db->shard];
   std::lock_guard<std::mutex> lk(ss.mu);
   /* the pull request's seq_no IS the follower's confirmed progress
-   * (mode-2 ack, replicated_db.cpp:452-456) */
-  if (since_seq > ss.acked_confirmed) ss.acked_confirmed = since_seq;
+   * (mode-2 ack); an OBSERVER's progress is ignored
+   * (replicated_db.cpp:452-456) */
+  if (requester_role == 0 && since_seq > ss.acked_confirmed)
+    ss.acked_confirmed = since_seq;
   *n_out = 0;
   if (!ss.log.empty() && since_seq + 1 < ss.log.front().base_seq) {
     /* reference analog: WAL no longer reaches back that far */
@@ -1227,7 +1229,8 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
     ss.cnt_served++;
     ss.cnt_out_bytes += ent.rep.size();
   }
-  if (*n_out > 0) { /* mode-1 ack: acked once sent (replicated_db.cpp:543-546) */
+  if (*n_out > 0 && requester_role == 0) {
+    /* mode-1 ack: acked once sent (replicated_db.cpp:543-546) */
     uint64_t last = out[*n_out - 1].seq;
     const LogEnt *le = nullptr;
     for (const LogEnt &ent : ss.log)

@@ -121,7 +121,7 @@ def load():
     lib.gra_get_updates.argtypes = [C.c_void_p, C.c_uint64, C.c_uint32,
                                     C.POINTER(GraServedUpdate),
                                     C.POINTER(C.c_uint32), C.c_char_p,
-                                    C.c_size_t]
+                                    C.c_size_t, C.c_int]
     lib.gra_wait_ack.argtypes = [C.c_void_p, C.c_uint64, C.c_int, C.c_int]
     lib.gra_db_counters.argtypes = [C.c_void_p, C.POINTER(GraDbCounters)]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
@@ -345,14 +345,17 @@ class Db:
         return self.lib.gra_wait_ack(self.h, seq, 1 if confirmed else 0,
                                      timeout_ms) == GRA_OK
 
-    def get_updates(self, since_seq, max_updates=50, cap=1 << 22):
+    def get_updates(self, since_seq, max_updates=50, cap=1 << 22,
+                    observer=False):
         """Leader serving: [(seq, ts, rep_bytes), ...] with
-        base seq > since_seq — the reference Update triple."""
+        base seq > since_seq — the reference Update triple. An observer's
+        request does not post an ACK (replicated_db.cpp:452-456)."""
         out = (GraServedUpdate * max_updates)()
         buf = C.create_string_buffer(cap)
         n = C.c_uint32()
         rc = self.lib.gra_get_updates(self.h, since_seq, max_updates, out,
-                                      C.byref(n), buf, cap)
+                                      C.byref(n), buf, cap,
+                                      1 if observer else 0)
         if rc != GRA_OK:
             raise RuntimeError(f"gra_get_updates rc={rc}: {last_error(self.lib)}")
         return [(out[i].seq, out[i].ts, buf.raw[out[i].off:out[i].off + out[i].len])

@@ -7,10 +7,14 @@ as the wiring template for a transport layer.
 """
 
 
-def pull_once(upstream_db, follower_db, max_updates=50):
+def pull_once(upstream_db, follower_db, max_updates=50, observer=False):
     """One pull round: returns number of updates applied."""
     since = follower_db.latest_seq()  # req.seq_no (replicated_db.cpp:317)
-    updates = upstream_db.get_updates(since, max_updates)
+    try:
+        updates = upstream_db.get_updates(since, max_updates,
+                                          observer=observer)
+    except TypeError:  # transports without the role parameter
+        updates = upstream_db.get_updates(since, max_updates)
     applied = 0
     for seq, ts, rep in updates:  # apply loop (replicated_db.cpp:369-383)
         if not follower_db.handle_replicate_response(rep, ts):
@@ -139,7 +143,8 @@ class Replicator:
         def loop():
             while not rs._stop.is_set():
                 try:
-                    n = pull_once(rs.upstream_db, rs.db)
+                    n = pull_once(rs.upstream_db, rs.db,
+                                  observer=(rs.role == OBSERVER))
                     if n:
                         self.engine.flush()
                         continue  # more may be pending: immediate re-pull (:430)

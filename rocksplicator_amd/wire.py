@@ -91,11 +91,14 @@ class UpdateServer:
                         if db is None:
                             self.request.sendall(encode_response([]))
                             continue
-                        ups = db.get_updates(seq_no, max_updates)
+                        obs = _role == ROLE_OBSERVER
+                        ups = db.get_updates(seq_no, max_updates,
+                                             observer=obs)
                         if not ups and max_wait_ms:
                             with outer._cond:  # long-poll
                                 outer._cond.wait(max_wait_ms / 1e3)
-                            ups = db.get_updates(seq_no, max_updates)
+                            ups = db.get_updates(seq_no, max_updates,
+                                                 observer=obs)
                         self.request.sendall(encode_response(ups))
                 except (ConnectionError, OSError):
                     pass
@@ -136,10 +139,11 @@ class RemoteUpstream:
         self._max_wait_ms = max_wait_ms
         self._mu = threading.Lock()
 
-    def get_updates(self, since_seq, max_updates=50):
+    def get_updates(self, since_seq, max_updates=50, observer=False):
         with self._mu:
-            self._sock.sendall(encode_request(since_seq, self._name,
-                                              self._max_wait_ms, max_updates))
+            self._sock.sendall(encode_request(
+                since_seq, self._name, self._max_wait_ms, max_updates,
+                ROLE_OBSERVER if observer else ROLE_FOLLOWER))
             return decode_response(self._sock)
 
     def latest_seq(self):

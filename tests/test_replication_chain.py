@@ -329,3 +329,22 @@ def test_concurrent_write_and_pull_soak(olib):
     assert int.from_bytes(ldb.get(b"ctr"), "little") == N
     leader.close()
     follower.close()
+
+
+def test_observer_pull_does_not_ack(olib):
+    """An OBSERVER's pull must not post the confirmed ACK
+    (replicated_db.cpp:452-456): a mode-2 wait stays unsatisfied by observer
+    progress but is satisfied by a follower's."""
+    e = ra.Engine(nshards=2, retain_log=1)
+    db = e.open(0)
+    rep = PyBatch().put(b"k", b"v").data()
+    seq = db.write_leader(rep)
+    # observer consumes the update — no ack
+    ups = db.get_updates(0, observer=True)
+    assert len(ups) == 1
+    assert not db.wait_ack(seq, confirmed=True, timeout_ms=100)
+    # follower request at seq (it already applied) — confirmed ack posts
+    ups2 = db.get_updates(seq, observer=False)
+    assert ups2 == []
+    assert db.wait_ack(seq, confirmed=True, timeout_ms=100)
+    e.close()
