@@ -34,8 +34,8 @@ HBM_PEAK_GBPS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md)
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--nshards", type=int, default=1024)
     p.add_argument("--key-len", type=int, default=16)
     p.add_argument("--val-len", type=int, default=1024)

@@ -348,3 +348,41 @@ def test_observer_pull_does_not_ack(olib):
     assert ups2 == []
     assert db.wait_ack(seq, confirmed=True, timeout_ms=100)
     e.close()
+
+
+def test_tcp_two_followers_one_leader(olib):
+    """Two followers pulling the same TCP leader concurrently (the
+    reference's tree topology, rocksdb_replicator_test.cpp:210-268)."""
+    from rocksplicator_amd import wire
+    from rocksplicator_amd.replicator import pull_once
+    leader = ra.Engine(nshards=1, retain_log=1)
+    f1 = ra.Engine(nshards=1)
+    f2 = ra.Engine(nshards=1)
+    srv = wire.UpdateServer()
+    ldb = leader.open(0)
+    srv.register("db", ldb)
+    r1 = wire.RemoteUpstream("127.0.0.1", srv.port, "db")
+    r2 = wire.RemoteUpstream("127.0.0.1", srv.port, "db")
+    d1, d2 = f1.open(0), f2.open(0)
+    for i in range(200):
+        ldb.write_leader(PyBatch().put(f"k{i % 23}".encode(),
+                                       f"v{i}".encode()).data())
+    import threading
+    def drain(remote, db, eng):
+        while pull_once(remote, db):
+            pass
+        eng.flush()
+    t1 = threading.Thread(target=drain, args=(r1, d1, f1))
+    t2 = threading.Thread(target=drain, args=(r2, d2, f2))
+    t1.start(); t2.start(); t1.join(60); t2.join(60)
+    assert d1.latest_seq() == d2.latest_seq() == 200
+    for i in range(23):
+        k = f"k{i}".encode()
+        assert d1.get(k) == d2.get(k) == ldb.get(k)
+    # both followers acked: mode-2 wait satisfied
+    assert ldb.wait_ack(200, confirmed=True, timeout_ms=100)
+    for x in (r1, r2):
+        x.close()
+    srv.close()
+    for e in (leader, f1, f2):
+        e.close()
