@@ -12,7 +12,7 @@ all: rocksplicator_amd/libgra.so oracle/libwb_oracle.so
 build:
 	mkdir -p build
 
-build/engine.o: $(CSRC)/engine.hip $(CSRC)/wb_format.h $(CSRC)/host_store.h include/rocksplicator_gpu.h | build
+build/engine.o: $(CSRC)/engine.hip $(CSRC)/wb_format.h $(CSRC)/host_store.h $(CSRC)/snappy.h include/rocksplicator_gpu.h | build
 	$(HIPCC) $(HIPFLAGS) -x hip -c $< -o $@
 
 build/host_store.o: $(CSRC)/host_store.cpp $(CSRC)/host_store.h $(CSRC)/wb_format.h | build
@@ -21,7 +21,7 @@ build/host_store.o: $(CSRC)/host_store.cpp $(CSRC)/host_store.h $(CSRC)/wb_forma
 build/builder.o: $(CSRC)/builder.cpp $(CSRC)/wb_format.h include/rocksplicator_gpu.h | build
 	$(HIPCC) $(HIPFLAGS) -c $< -o $@
 
-build/gen.o: $(CSRC)/gen.cpp $(CSRC)/wb_format.h include/rocksplicator_gpu.h | build
+build/gen.o: $(CSRC)/gen.cpp $(CSRC)/wb_format.h $(CSRC)/snappy.h include/rocksplicator_gpu.h | build
 	$(HIPCC) $(HIPFLAGS) -c $< -o $@
 
 rocksplicator_amd/libgra.so: $(OBJS)

@@ -456,7 +456,6 @@ struct GraEngine {
   /* device store */
   uint8_t *d_store = nullptr;
   uint64_t *d_cursor = nullptr;
-  uint64_t cursor_seen = 0; /* host view (from ingested ticks) */
   /* tick scratch */
   uint32_t max_upd;          /* max updates per tick */
   uint32_t task_cap;
