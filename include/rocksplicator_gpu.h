@@ -114,6 +114,26 @@ int gra_get(GraDb *db, const void *key, size_t klen, void *buf, size_t cap,
  * this, gra_latest_seq and gra_get are linearizable per shard. */
 int gra_flush(GraEngine *e);
 
+/* Batched point reads served FROM THE DEVICE STORE (followers serve reads
+ * in rocksplicator deployments — ApplicationDB::Get routes to the local db;
+ * here the memtable lives in HBM, so the search runs there too: one block
+ * per query scanning the shard's runs newest->oldest). Each result value is
+ * written at valbuf[q*val_stride] (vlen capped at val_stride). A query
+ * whose outcome needs merge-operand folding reports GRA_GET_NEEDS_HOST and
+ * is answered by the host path (gra_get) instead. Call gra_flush first for
+ * linearizable results. */
+enum { GRA_GET_FOUND = 0, GRA_GET_MISS = 1, GRA_GET_NEEDS_HOST = 2 };
+typedef struct {
+  uint32_t off, len; /* key slice in keybuf */
+} GraKeyRef;
+typedef struct {
+  uint32_t status; /* GRA_GET_* */
+  uint32_t vlen;
+} GraGetResult;
+int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
+                 const uint8_t *keybuf, size_t keybuf_len, uint8_t *valbuf,
+                 uint32_t val_stride, GraGetResult *out);
+
 /* ---------------- leader update-serving ----------------
  * ≅ ReplicatedDB::handleReplicateRequest batch read-out (replicated_db.cpp:
  * 435-575): serve retained batches with base seq > since_seq, up to

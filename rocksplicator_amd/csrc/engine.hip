@@ -412,6 +412,122 @@ __global__ void k_rundesc(const GroupDesc *__restrict__ groups, uint32_t ngroups
   out[g] = rd;
 }
 
+/* ---------------- device read serving (multiget) ----------------
+ * One block per query: scan the shard's runs newest->oldest; per matching
+ * entry track the max-seq terminator (Put/Delete/SingleDelete), whether any
+ * Merge sits above it (=> host fold), and the max covering range-tombstone
+ * seq. Followers serve reads in rocksplicator deployments; with the
+ * memtable resident in HBM the lookup runs there too. */
+struct RunView {
+  uint64_t hdr_off, payload_off;
+  uint32_t n_entries, pay_rel_base;
+};
+
+__device__ inline int dev_memcmp(const uint8_t *a, const uint8_t *b, uint32_t n) {
+  for (uint32_t i = 0; i < n; i++) {
+    if (a[i] != b[i]) return a[i] < b[i] ? -1 : 1;
+  }
+  return 0;
+}
+
+__global__ void k_multiget(const uint8_t *__restrict__ store,
+                           const RunView *__restrict__ runs, uint32_t nruns,
+                           const GraKeyRef *__restrict__ keys,
+                           const uint8_t *__restrict__ keybuf, uint32_t nq,
+                           uint8_t *__restrict__ valbuf, uint32_t val_stride,
+                           GraGetResult *__restrict__ out) {
+  __shared__ uint64_t sh_term_seq[256];
+  __shared__ uint64_t sh_term_ref[256]; /* (run<<32)|entry, ~0 = none */
+  __shared__ uint64_t sh_merge_seq[256];
+  __shared__ uint64_t sh_rd_seq[256];
+  uint32_t q = blockIdx.x;
+  if (q >= nq) return;
+  const uint8_t *key = keybuf + keys[q].off;
+  uint32_t klen = keys[q].len;
+  uint64_t term_seq = 0, term_ref = ~0ULL, merge_seq = 0, rd_seq = 0;
+  for (uint32_t r = 0; r < nruns; r++) {
+    RunView rv = runs[r];
+    const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+    const uint8_t *pay = store + rv.payload_off;
+    for (uint32_t i = threadIdx.x; i < rv.n_entries; i += blockDim.x) {
+      wb::RecHdr h = hdrs[i];
+      uint32_t rel = h.kv_off - rv.pay_rel_base;
+      if (h.type == wb::kRangeDeletion) {
+        const uint8_t *b = pay + rel, *e2 = pay + rel + h.key_len;
+        uint32_t bl = h.key_len, el = h.val_len;
+        int c1 = dev_memcmp(b, key, bl < klen ? bl : klen);
+        if (c1 > 0 || (c1 == 0 && bl > klen)) continue;
+        int c2 = dev_memcmp(key, e2, klen < el ? klen : el);
+        if (c2 > 0 || (c2 == 0 && klen >= el)) continue;
+        if (h.seq > rd_seq) rd_seq = h.seq;
+        continue;
+      }
+      if (h.key_len != klen || dev_memcmp(pay + rel, key, klen) != 0) continue;
+      if (h.type == wb::kMerge) {
+        if (h.seq > merge_seq) merge_seq = h.seq;
+      } else if (h.seq > term_seq) {
+        term_seq = h.seq;
+        term_ref = ((uint64_t)r << 32) | i;
+      }
+    }
+  }
+  uint32_t tid = threadIdx.x;
+  sh_term_seq[tid] = term_seq;
+  sh_term_ref[tid] = term_ref;
+  sh_merge_seq[tid] = merge_seq;
+  sh_rd_seq[tid] = rd_seq;
+  __syncthreads();
+  for (uint32_t ofs = blockDim.x / 2; ofs > 0; ofs >>= 1) {
+    if (tid < ofs) {
+      if (sh_term_seq[tid + ofs] > sh_term_seq[tid]) {
+        sh_term_seq[tid] = sh_term_seq[tid + ofs];
+        sh_term_ref[tid] = sh_term_ref[tid + ofs];
+      }
+      if (sh_merge_seq[tid + ofs] > sh_merge_seq[tid])
+        sh_merge_seq[tid] = sh_merge_seq[tid + ofs];
+      if (sh_rd_seq[tid + ofs] > sh_rd_seq[tid])
+        sh_rd_seq[tid] = sh_rd_seq[tid + ofs];
+    }
+    __syncthreads();
+  }
+  uint64_t T = sh_term_seq[0], M = sh_merge_seq[0], RD = sh_rd_seq[0];
+  uint64_t ref = sh_term_ref[0];
+  uint64_t floor_seq = T > RD ? T : RD;
+  if (M > floor_seq) { /* live merge operands: fold on the host */
+    if (tid == 0) {
+      out[q].status = GRA_GET_NEEDS_HOST;
+      out[q].vlen = 0;
+    }
+    return;
+  }
+  if (T == 0 || T <= RD || ref == ~0ULL) {
+    if (tid == 0) {
+      out[q].status = GRA_GET_MISS;
+      out[q].vlen = 0;
+    }
+    return;
+  }
+  RunView rv = runs[ref >> 32];
+  const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+  wb::RecHdr h = hdrs[(uint32_t)ref];
+  if (h.type != wb::kValue) { /* Delete/SingleDelete terminator */
+    if (tid == 0) {
+      out[q].status = GRA_GET_MISS;
+      out[q].vlen = 0;
+    }
+    return;
+  }
+  uint32_t vlen = h.val_len < val_stride ? h.val_len : val_stride;
+  const uint8_t *src = store + rv.payload_off + (h.kv_off - rv.pay_rel_base) +
+                       h.key_len;
+  uint8_t *dst = valbuf + (size_t)q * val_stride;
+  for (uint32_t b = tid; b < vlen; b += blockDim.x) dst[b] = src[b];
+  if (tid == 0) {
+    out[q].status = GRA_GET_FOUND;
+    out[q].vlen = h.val_len; /* true length (caller sees truncation) */
+  }
+}
+
 /* ---------------- host engine ---------------- */
 
 struct Stats {
@@ -1309,6 +1425,90 @@ int gra_get(GraDb *db, const void *key, size_t klen, void *buf, size_t cap,
   memcpy(buf, out.data(), out.size());
   if (vlen) *vlen = out.size();
   return GRA_OK;
+}
+
+int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
+                 const uint8_t *keybuf, size_t keybuf_len, uint8_t *valbuf,
+                 uint32_t val_stride, GraGetResult *out) {
+  GraEngine *e = db->e;
+  if (nq == 0) return GRA_OK;
+  ShardState &ss = e->shards[db->shard];
+  /* snapshot the device-resident run list (newest->oldest for the kernel);
+   * host-origin runs (leader writes) have no device bytes — fall back to the
+   * host path for the whole call when any exist (mixed shards are the
+   * exception, not the serving case) */
+  std::vector<RunView> views;
+  {
+    std::lock_guard<std::mutex> lk(ss.mu);
+    views.reserve(ss.runs.size());
+    for (auto it = ss.runs.rbegin(); it != ss.runs.rend(); ++it) {
+      const Run &r = **it;
+      if (r.n_entries == 0) continue;
+      if (r.hdr_cur == UINT64_MAX) { /* host-origin run */
+        views.clear();
+        break;
+      }
+      views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
+    }
+    if (views.empty() && !ss.runs.empty()) {
+      /* host fallback: answer every query via gra_get */
+      for (uint32_t q = 0; q < nq; q++) out[q].status = GRA_GET_NEEDS_HOST;
+      return GRA_OK;
+    }
+  }
+  if (views.empty()) {
+    for (uint32_t q = 0; q < nq; q++) {
+      out[q].status = GRA_GET_MISS;
+      out[q].vlen = 0;
+    }
+    return GRA_OK;
+  }
+  /* device staging for this call (synchronous; serving path would keep a
+   * persistent device run index — round-2 refinement) */
+  RunView *d_runs = nullptr;
+  GraKeyRef *d_keys = nullptr;
+  uint8_t *d_keybuf = nullptr, *d_valbuf = nullptr;
+  GraGetResult *d_out = nullptr;
+  auto cleanup = [&] {
+    if (d_runs) (void)hipFree(d_runs);
+    if (d_keys) (void)hipFree(d_keys);
+    if (d_keybuf) (void)hipFree(d_keybuf);
+    if (d_valbuf) (void)hipFree(d_valbuf);
+    if (d_out) (void)hipFree(d_out);
+  };
+  size_t vb = (size_t)nq * val_stride;
+  if (hipMalloc(&d_runs, views.size() * sizeof(RunView)) != hipSuccess ||
+      hipMalloc(&d_keys, nq * sizeof(GraKeyRef)) != hipSuccess ||
+      hipMalloc(&d_keybuf, keybuf_len + 16) != hipSuccess ||
+      hipMalloc(&d_valbuf, vb + 16) != hipSuccess ||
+      hipMalloc(&d_out, nq * sizeof(GraGetResult)) != hipSuccess) {
+    cleanup();
+    g_err = "gra_multiget: allocation failed";
+    return GRA_ERR;
+  }
+  int rc = GRA_ERR;
+  do {
+    if (hipMemcpy(d_runs, views.data(), views.size() * sizeof(RunView),
+                  hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(d_keys, keys, nq * sizeof(GraKeyRef),
+                  hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(d_keybuf, keybuf, keybuf_len, hipMemcpyHostToDevice) !=
+            hipSuccess)
+      break;
+    hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream, e->d_store,
+                       d_runs, (uint32_t)views.size(), d_keys, d_keybuf, nq,
+                       d_valbuf, val_stride, d_out);
+    if (hipGetLastError() != hipSuccess) break;
+    if (hipStreamSynchronize(e->stream) != hipSuccess) break;
+    if (hipMemcpy(out, d_out, nq * sizeof(GraGetResult),
+                  hipMemcpyDeviceToHost) != hipSuccess ||
+        hipMemcpy(valbuf, d_valbuf, vb, hipMemcpyDeviceToHost) != hipSuccess)
+      break;
+    rc = GRA_OK;
+  } while (0);
+  if (rc != GRA_OK) g_err = "gra_multiget: device op failed";
+  cleanup();
+  return rc;
 }
 
 int gra_pin_alloc(GraEngine *e, size_t bytes, uint8_t **ptr) {

@@ -38,6 +38,14 @@ class GraEngineOpts(C.Structure):
     ]
 
 
+class GraKeyRef(C.Structure):
+    _fields_ = [("off", C.c_uint32), ("len", C.c_uint32)]
+
+
+class GraGetResult(C.Structure):
+    _fields_ = [("status", C.c_uint32), ("vlen", C.c_uint32)]
+
+
 class GraDbCounters(C.Structure):
     _fields_ = [(n, C.c_uint64) for n in (
         "updates_applied", "in_bytes", "apply_failures", "updates_served",
@@ -124,6 +132,9 @@ def load():
                                     C.c_size_t, C.c_int]
     lib.gra_wait_ack.argtypes = [C.c_void_p, C.c_uint64, C.c_int, C.c_int]
     lib.gra_db_counters.argtypes = [C.c_void_p, C.POINTER(GraDbCounters)]
+    lib.gra_multiget.argtypes = [C.c_void_p, C.c_uint32, C.POINTER(GraKeyRef),
+                                 C.c_char_p, C.c_size_t, C.c_char_p,
+                                 C.c_uint32, C.POINTER(GraGetResult)]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
     lib.gra_pin_free.argtypes = [C.c_void_p, C.POINTER(C.c_uint8)]
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
@@ -331,6 +342,37 @@ class Db:
         if rc != GRA_OK:
             raise RuntimeError(f"gra_get rc={rc}: {last_error(self.lib)}")
         return buf.raw[: vlen.value]
+
+    def multiget(self, keys, val_stride=4096):
+        """Batched point reads served from the device store (one block per
+        query); merge-folding queries transparently fall back to the host
+        path. Returns a list of values (None = miss)."""
+        nq = len(keys)
+        if nq == 0:
+            return []
+        keybuf = b"".join(keys)
+        refs = (GraKeyRef * nq)()
+        off = 0
+        for i, k in enumerate(keys):
+            refs[i] = GraKeyRef(off, len(k))
+            off += len(k)
+        valbuf = C.create_string_buffer(nq * val_stride)
+        out = (GraGetResult * nq)()
+        rc = self.lib.gra_multiget(self.h, nq, refs, keybuf, len(keybuf),
+                                   valbuf, val_stride, out)
+        if rc != GRA_OK:
+            raise RuntimeError(f"gra_multiget rc={rc}: {last_error(self.lib)}")
+        res = []
+        for i in range(nq):
+            st = out[i].status
+            if st == 1:  # miss
+                res.append(None)
+            elif st == 0:  # found
+                assert out[i].vlen <= val_stride, "value exceeds stride"
+                res.append(valbuf.raw[i * val_stride:i * val_stride + out[i].vlen])
+            else:  # needs host (merge fold / host-origin runs)
+                res.append(self.get(keys[i]))
+        return res
 
     def counters(self):
         """Per-db stats ≅ the reference's per-db counter fan-out

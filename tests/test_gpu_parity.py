@@ -601,3 +601,40 @@ def test_corruption_decision_matrix(olib):
         k = f"k{i}".encode()
         assert db.get(k) == ost.get(0, k)
     e.close()
+
+
+def test_multiget_device_reads(olib):
+    """Batched point reads served by k_multiget from the device store ==
+    host-path gra_get == oracle, across puts/overwrites/deletes/range
+    tombstones/merges (merge queries route to the host fold)."""
+    nshards, nupd = 16, 30000
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=200, kind=2,
+                                       key_space=1 << 12, seed=808)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd, merge_op=0)
+    e = ra.Engine(nshards=nshards, merge_op=0)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    rep.tick(0, nupd)
+    rep.sync()
+    keys = collect_keys(olib, raw, descs, 8000, per_shard_cap=64)
+    for s in range(0, nshards, 3):
+        db = e.open(s)
+        ks = keys.get(s, []) + [b"definitely-missing-key-0123"]
+        got = db.multiget(ks)
+        for k, v in zip(ks, got):
+            assert v == ost.get(s, k), (s, k.hex())
+        db.close()
+    # range tombstone + merge interplay through the device path
+    db = e.open(0)
+    assert db.handle_replicate_response(PyBatch().put(b"rk1", b"a").put(b"rk2", b"b").data())
+    assert db.handle_replicate_response(PyBatch().delete_range(b"rk1", b"rk2").data())
+    assert db.handle_replicate_response(PyBatch().merge(b"rk3", b"m1").data())
+    e.flush()
+    ost.apply(0, PyBatch().put(b"rk1", b"a").put(b"rk2", b"b").data())
+    ost.apply(0, PyBatch().delete_range(b"rk1", b"rk2").data())
+    ost.apply(0, PyBatch().merge(b"rk3", b"m1").data())
+    got = db.multiget([b"rk1", b"rk2", b"rk3"])
+    assert got[0] is None and got[1] == b"b" and got[2] == b"m1"
+    assert [ost.get(0, k) for k in (b"rk1", b"rk2", b"rk3")] == got
+    e.close()
