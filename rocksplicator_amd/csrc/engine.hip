@@ -276,7 +276,9 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = wb::base_tag(r.tag);
     h.flags = cf4 ? 1 : 0;
-    h._pad = 0;
+    /* stored key = [cf?4B][key bytes]; prefix covers the stored form */
+    h.kpref = cf4 ? r.cf_id
+                  : wb::key_prefix4(blobs + d.off + r.key_off, r.key_len);
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -444,6 +446,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   if (q >= nq) return;
   const uint8_t *key = keybuf + keys[q].off;
   uint32_t klen = keys[q].len;
+  uint32_t qpref = wb::key_prefix4(key, klen);
   uint64_t term_seq = 0, term_ref = ~0ULL, merge_seq = 0, rd_seq = 0;
   for (uint32_t r = 0; r < nruns; r++) {
     RunView rv = runs[r];
@@ -462,7 +465,9 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
         if (h.seq > rd_seq) rd_seq = h.seq;
         continue;
       }
-      if (h.key_len != klen || dev_memcmp(pay + rel, key, klen) != 0) continue;
+      if (h.key_len != klen || h.kpref != qpref ||
+          dev_memcmp(pay + rel, key, klen) != 0)
+        continue;
       if (h.type == wb::kMerge) {
         if (h.seq > merge_seq) merge_seq = h.seq;
       } else if (h.seq > term_seq) {
@@ -606,6 +611,14 @@ struct GraEngine {
   Slot slots[kSlots];
   std::deque<TickRec> pending;
   std::vector<hipEvent_t> event_pool;
+  /* multiget staging cache (grow-only; guarded by mu) */
+  struct MgCache {
+    RunView *d_runs = nullptr;
+    GraKeyRef *d_keys = nullptr;
+    uint8_t *d_keybuf = nullptr, *d_valbuf = nullptr;
+    GraGetResult *d_out = nullptr;
+    size_t runs_cap = 0, keys_cap = 0, keybuf_cap = 0, valbuf_cap = 0, out_cap = 0;
+  } mg;
   /* shards */
   std::vector<ShardState> shards;
   Stats stats;
@@ -750,6 +763,9 @@ GraEngine::~GraEngine() {
     if (s.h_err) (void)hipHostFree(s.h_err);
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
+  for (void *p : {(void *)mg.d_runs, (void *)mg.d_keys, (void *)mg.d_keybuf,
+                  (void *)mg.d_valbuf, (void *)mg.d_out})
+    if (p) (void)hipFree(p);
   for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_totals,
                   (void *)d_partial, (void *)d_bsums, (void *)d_reccache,
                   (void *)d_tasks, (void *)d_place, (void *)d_groups,
@@ -1463,29 +1479,33 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
     }
     return GRA_OK;
   }
-  /* device staging for this call (synchronous; serving path would keep a
-   * persistent device run index — round-2 refinement) */
-  RunView *d_runs = nullptr;
-  GraKeyRef *d_keys = nullptr;
-  uint8_t *d_keybuf = nullptr, *d_valbuf = nullptr;
-  GraGetResult *d_out = nullptr;
-  auto cleanup = [&] {
-    if (d_runs) (void)hipFree(d_runs);
-    if (d_keys) (void)hipFree(d_keys);
-    if (d_keybuf) (void)hipFree(d_keybuf);
-    if (d_valbuf) (void)hipFree(d_valbuf);
-    if (d_out) (void)hipFree(d_out);
+  /* grow-only device staging cached on the engine (serialized with other
+   * engine ops by mu; a serving deployment would shard this per stream) */
+  std::lock_guard<std::mutex> lk_engine(e->mu);
+  auto grow = [](auto **p, size_t *cap, size_t need) {
+    if (*cap >= need) return true;
+    if (*p) (void)hipFree(*p);
+    *p = nullptr;
+    *cap = 0;
+    size_t want = need + need / 2 + 64;
+    if (hipMalloc(p, want) != hipSuccess) return false;
+    *cap = want;
+    return true;
   };
   size_t vb = (size_t)nq * val_stride;
-  if (hipMalloc(&d_runs, views.size() * sizeof(RunView)) != hipSuccess ||
-      hipMalloc(&d_keys, nq * sizeof(GraKeyRef)) != hipSuccess ||
-      hipMalloc(&d_keybuf, keybuf_len + 16) != hipSuccess ||
-      hipMalloc(&d_valbuf, vb + 16) != hipSuccess ||
-      hipMalloc(&d_out, nq * sizeof(GraGetResult)) != hipSuccess) {
-    cleanup();
+  auto &mg = e->mg;
+  if (!grow(&mg.d_runs, &mg.runs_cap, views.size() * sizeof(RunView)) ||
+      !grow(&mg.d_keys, &mg.keys_cap, nq * sizeof(GraKeyRef)) ||
+      !grow(&mg.d_keybuf, &mg.keybuf_cap, keybuf_len + 16) ||
+      !grow(&mg.d_valbuf, &mg.valbuf_cap, vb + 16) ||
+      !grow(&mg.d_out, &mg.out_cap, nq * sizeof(GraGetResult))) {
     g_err = "gra_multiget: allocation failed";
     return GRA_ERR;
   }
+  RunView *d_runs = mg.d_runs;
+  GraKeyRef *d_keys = mg.d_keys;
+  uint8_t *d_keybuf = mg.d_keybuf, *d_valbuf = mg.d_valbuf;
+  GraGetResult *d_out = mg.d_out;
   int rc = GRA_ERR;
   do {
     if (hipMemcpy(d_runs, views.data(), views.size() * sizeof(RunView),
@@ -1507,7 +1527,6 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
     rc = GRA_OK;
   } while (0);
   if (rc != GRA_OK) g_err = "gra_multiget: device op failed";
-  cleanup();
   return rc;
 }
 

@@ -223,8 +223,15 @@ struct RecHdr {
   uint16_t key_len;  /* includes 4-byte cf prefix when cf_id != 0 */
   uint8_t type;      /* base tag */
   uint8_t flags;     /* bit0: cf-prefixed key */
-  uint32_t _pad;
+  uint32_t kpref;    /* first min(4,key_len) key bytes, zero-padded — lets
+                        point lookups filter without touching the payload */
 };
+
+WB_HD uint32_t key_prefix4(const uint8_t *key, uint32_t klen) {
+  uint32_t p = 0;
+  for (uint32_t i = 0; i < 4 && i < klen; i++) p |= (uint32_t)key[i] << (8 * i);
+  return p;
+}
 static_assert(sizeof(RecHdr) == 24, "RecHdr must be 24 bytes");
 
 } /* namespace wb */

@@ -362,6 +362,7 @@ class Db:
                                    valbuf, val_stride, out)
         if rc != GRA_OK:
             raise RuntimeError(f"gra_multiget rc={rc}: {last_error(self.lib)}")
+        raw = valbuf.raw  # one copy; per-slice .raw would copy 2.5MB each
         res = []
         for i in range(nq):
             st = out[i].status
@@ -369,7 +370,7 @@ class Db:
                 res.append(None)
             elif st == 0:  # found
                 assert out[i].vlen <= val_stride, "value exceeds stride"
-                res.append(valbuf.raw[i * val_stride:i * val_stride + out[i].vlen])
+                res.append(raw[i * val_stride:i * val_stride + out[i].vlen])
             else:  # needs host (merge fold / host-origin runs)
                 res.append(self.get(keys[i]))
         return res
