@@ -638,3 +638,22 @@ def test_multiget_device_reads(olib):
     assert got[0] is None and got[1] == b"b" and got[2] == b"m1"
     assert [ost.get(0, k) for k in (b"rk1", b"rk2", b"rk3")] == got
     e.close()
+
+
+def test_oversize_batch_rejected(olib):
+    """A batch with more records than max_wb_records (DoS guard) is rejected
+    like corruption: shard poisons, seq rolls back, clean recovery."""
+    e = ra.Engine(nshards=1)  # default cap 1024 records
+    db = e.open(0)
+    big = PyBatch()
+    for i in range(1100):
+        big.put(f"k{i}".encode(), b"v")
+    assert db.handle_replicate_response(big.data())  # accepted (async)
+    e.flush()
+    assert db.latest_seq() == 0  # rolled back
+    assert not db.handle_replicate_response(PyBatch().put(b"a", b"b").data())
+    assert db.handle_replicate_response(PyBatch().put(b"a", b"b").data())
+    e.flush()
+    assert db.latest_seq() == 1 and db.get(b"a") == b"b"
+    # oracle has no record cap: this is an engine-level operational guard
+    e.close()
