@@ -1742,6 +1742,11 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
     g_err = "replay window out of range";
     return GRA_ERR;
   }
+  if (r->snappy || r->h_arena == nullptr) {
+    g_err = "tick_h2d needs a host-resident uncompressed arena "
+            "(use gra_replay_tick for snappy/device uploads)";
+    return GRA_ERR;
+  }
   /* window blobs must be contiguous in the arena (generator layout) */
   uint64_t lo = r->descs[first].off;
   uint64_t hi = r->descs[first + n - 1].off + r->descs[first + n - 1].len;

@@ -127,7 +127,7 @@ __device__ void copy_v3(uint8_t *dst, const uint8_t *src, uint32_t n, uint32_t l
   }
 }
 
-template <int V, int G>
+template <int V, int G, int STEP = 1, int OFS = 0>
 __global__ void __launch_bounds__(256) k_bench(const uint8_t *__restrict__ src,
                                                uint8_t *__restrict__ dst,
                                                const Task *__restrict__ tasks,
@@ -135,7 +135,7 @@ __global__ void __launch_bounds__(256) k_bench(const uint8_t *__restrict__ src,
   uint32_t lane = threadIdx.x & (G - 1);
   uint32_t g = (blockIdx.x * blockDim.x + threadIdx.x) / G;
   uint32_t ngroups = (gridDim.x * blockDim.x) / G;
-  for (uint32_t t = g; t < ntasks; t += ngroups) {
+  for (uint32_t t = g * STEP + OFS; t < ntasks; t += ngroups * STEP) {
     Task tk = tasks[t];
     if (V == 0) copy_v0<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
     if (V == 1) copy_v1<G>(dst + tk.dst_off, src + tk.src_off, tk.nbytes, lane);
@@ -205,5 +205,21 @@ int main(int argc, char **argv) {
   RUNG(2, 32, 1024); RUNG(2, 32, 4096); RUNG(2, 32, 8192);
   RUNG(3, 32, 1024); RUNG(3, 32, 4096); RUNG(3, 32, 8192);
   RUNG(3, 16, 4096);
+  /* split streams: value-only tasks (odd) at g32, key-only (even) at g16 —
+   * does de-mixing the task sizes in a wave buy anything? */
+  run("vals/g32", [&] {
+    hipLaunchKernelGGL((k_bench<2, 32, 2, 1>), dim3(2048), blk, 0, 0, d_src,
+                       d_dst, d_tasks, nt);
+  });
+  run("keys/g16", [&] {
+    hipLaunchKernelGGL((k_bench<2, 16, 2, 0>), dim3(2048), blk, 0, 0, d_src,
+                       d_dst, d_tasks, nt);
+  });
+  run("split-both", [&] {
+    hipLaunchKernelGGL((k_bench<2, 32, 2, 1>), dim3(2048), blk, 0, 0, d_src,
+                       d_dst, d_tasks, nt);
+    hipLaunchKernelGGL((k_bench<2, 16, 2, 0>), dim3(2048), blk, 0, 0, d_src,
+                       d_dst, d_tasks, nt);
+  });
   return 0;
 }
