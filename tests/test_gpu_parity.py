@@ -657,3 +657,34 @@ def test_oversize_batch_rejected(olib):
     assert db.latest_seq() == 1 and db.get(b"a") == b"b"
     # oracle has no record cap: this is an engine-level operational guard
     e.close()
+
+
+def test_tick_h2d_rejects_snappy_and_device_uploads(olib):
+    import ctypes as CT
+    plib = ra.load()
+    arena, used, descs = ra.gen_stream(nshards=2, n_updates=100, key_len=16,
+                                       val_len=64, kind=2, seed=1,
+                                       compressible=1)
+    raw = bytes(arena)[:used]
+    comp = CT.create_string_buffer(used + used // 4 + 64 * 100)
+    cdescs = (ra.ffi.GraUpdateDesc * 100)()
+    ulens = []
+    off = 0
+    for i in range(100):
+        d = descs[i]
+        blob = raw[d.off:d.off + d.len]
+        cap = len(blob) + len(blob) // 6 + 64
+        tmp = CT.create_string_buffer(cap)
+        clen = plib.gra_snappy_compress(blob, len(blob), tmp, cap)
+        CT.memmove(CT.byref(comp, off), tmp, clen)
+        cdescs[i] = ra.ffi.GraUpdateDesc(d.shard, clen, off, 0)
+        ulens.append(d.len)
+        off += clen
+    e = ra.Engine(nshards=2)
+    rep = e.upload_snappy(CT.cast(comp, CT.POINTER(CT.c_uint8)), off,
+                          cdescs, 100, ulens)
+    with pytest.raises(RuntimeError, match="tick_h2d"):
+        rep.tick_h2d(0, 50)
+    rep.tick(0, 100)  # the supported path still works
+    rep.sync()
+    e.close()
