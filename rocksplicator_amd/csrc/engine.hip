@@ -600,6 +600,8 @@ struct GraEngine {
     alignas(64) std::atomic<uint32_t> nslots{0};
     alignas(64) std::atomic<int> writers{0};
     alignas(64) std::atomic<bool> closed{false};
+    alignas(64) std::atomic<uint32_t> staged{0}; /* successful desc writes
+        (cross-checked against the tick build's count) */
     uint64_t epoch = 0; /* bumped at every swap: invalidates thread chunks */
     hipEvent_t free_ev = nullptr; /* recorded after this buffer's H2D */
   };
@@ -1081,6 +1083,11 @@ int GraEngine::stream_tick_locked() {
   HIP_TRY(hipEventSynchronize(next->free_ev));
   next->pos.store(0, std::memory_order_relaxed);
   next->nslots.store(0, std::memory_order_relaxed);
+  next->staged.store(0, std::memory_order_relaxed);
+  next->epoch++; /* invalidates every thread-local chunk of the previous
+                    generation (sequenced before the release publish below;
+                    without this, stale chunks write into recycled slots —
+                    lost updates under flush churn) */
   next->closed.store(false, std::memory_order_release);
   /* swap: new writers land in `next`; then quiesce `old` */
   old->closed.store(true, std::memory_order_release);
@@ -1097,6 +1104,14 @@ int GraEngine::stream_tick_locked() {
     if (old->descs[i].len) cnt[old->descs[i].shard + 1]++;
   for (uint32_t s = 0; s < opts.nshards; s++) cnt[s + 1] += cnt[s];
   uint32_t n = cnt[opts.nshards];
+  { /* invariant: every successful HandleReplicateResponse of this buffer
+       generation must surface exactly once in the tick build */
+    uint32_t expect = old->staged.load(std::memory_order_relaxed);
+    if (n != expect)
+      fprintf(stderr,
+              "[gra] BUG: tick build found %u staged updates, expected %u\n",
+              n, expect);
+  }
   if (n == 0) return GRA_OK;
   std::vector<UpdDesc> ud(n);
   std::vector<GroupDesc> groups;
@@ -1206,26 +1221,19 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     if (len < wb::kHeaderBytes) return 0;
     count = wb::fixed32_le(rep + 8);
     base = ss.next_seq;
-    ss.next_seq += count;
-    /* ts beyond this point is wire metadata only: the LogData(ts) trailer
-     * the reference appends (rocksdb_wrapper.cpp:19-20) is WAL-only and
-     * never reaches the memtable. Chained nodes re-serve with this ts. */
-    if (e->opts.retain_log) retain_locked(e, ss, base, count, ts, rep, len);
-    ss.cnt_updates++;
-    ss.cnt_in_bytes += len; /* ≅ kReplicatorInBytes (replicated_db.cpp:409) */
-    if (ts != 0) { /* ≅ kReplicatorLatency (replicated_db.cpp:370-374) */
-      int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
-          std::chrono::system_clock::now().time_since_epoch()).count());
-      ss.lat_sum_ms += (uint64_t)(now_ms > ts ? now_ms - ts : 0);
-      ss.lat_n++;
-    }
+    ss.next_seq += count; /* optimistic; rolled back on ANY failure below */
   }
+  auto rollback = [&] {
+    std::lock_guard<std::mutex> lk(ss.mu);
+    /* per-shard calls are sequential (the seam contract), so the optimistic
+     * advance is still the tail and can be undone */
+    if (ss.next_seq == base + count) ss.next_seq = base;
+  };
   if (len + 16 > e->opts.staging_bytes) {
     /* a single Update larger than the staging buffer cannot be staged;
      * refuse it (the reference's responses are bounded by max_updates x
      * batch size — this is a misconfiguration, not a data error) */
-    std::lock_guard<std::mutex> lk2(ss.mu);
-    ss.next_seq = base; /* roll the optimistic assignment back */
+    rollback();
     return 0;
   }
   /* lock-free staging with per-thread CHUNK reservation: a thread grabs a
@@ -1243,7 +1251,8 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
   thread_local Chunk ck;
   constexpr uint64_t kChunkBytes = 256 << 10;
   constexpr uint32_t kChunkSlots = 128;
-  for (int attempt = 0; attempt < 4000; attempt++) {
+  bool staged = false;
+  for (int attempt = 0; attempt < 100000 && !staged; attempt++) {
     GraEngine::StageBuf *sb = e->cur_stage.load(std::memory_order_acquire);
     sb->writers.fetch_add(1, std::memory_order_acq_rel);
     if (sb->closed.load(std::memory_order_acquire)) {
@@ -1259,15 +1268,17 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
       uint32_t slot = sb->nslots.fetch_add(kChunkSlots, std::memory_order_relaxed);
       if (off + want + 16 > e->opts.staging_bytes ||
           slot + kChunkSlots > e->max_upd) {
-        /* buffer full: abandon (slots pre-zeroed below never happened —
-         * zero whatever part of the slot run is in range) */
+        /* buffer full: abandon the reserved slots (len = 0) and tick */
         for (uint32_t i = slot; i < slot + kChunkSlots && i < e->max_upd; i++)
           sb->descs[i].len = 0;
         ck.sb = nullptr;
         sb->writers.fetch_sub(1, std::memory_order_acq_rel);
         std::lock_guard<std::mutex> lk(e->mu);
         if (e->cur_stage.load(std::memory_order_acquire) == sb) {
-          if (e->stream_tick_locked() != GRA_OK) return 0;
+          if (e->stream_tick_locked() != GRA_OK) {
+            rollback();
+            return 0;
+          }
         }
         continue;
       }
@@ -1285,11 +1296,30 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     ck.used += len;
     ck.slot++;
     ck.slots_left--;
+    sb->staged.fetch_add(1, std::memory_order_relaxed);
     sb->writers.fetch_sub(1, std::memory_order_release);
-    return 1;
+    staged = true;
   }
-  g_err = "staging contention: could not reserve a slot";
-  return 0;
+  if (!staged) {
+    g_err = "staging contention: could not reserve a slot";
+    rollback();
+    return 0;
+  }
+  /* success: bookkeeping + retention (AFTER staging so a failure above
+   * never leaves a retained-log entry for an un-applied batch) */
+  {
+    std::lock_guard<std::mutex> lk(ss.mu);
+    if (e->opts.retain_log) retain_locked(e, ss, base, count, ts, rep, len);
+    ss.cnt_updates++;
+    ss.cnt_in_bytes += len; /* ≅ kReplicatorInBytes (replicated_db.cpp:409) */
+    if (ts != 0) { /* ≅ kReplicatorLatency (replicated_db.cpp:370-374) */
+      int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
+          std::chrono::system_clock::now().time_since_epoch()).count());
+      ss.lat_sum_ms += (uint64_t)(now_ms > ts ? now_ms - ts : 0);
+      ss.lat_n++;
+    }
+  }
+  return 1;
 }
 
 uint64_t gra_latest_seq(GraDb *db) {

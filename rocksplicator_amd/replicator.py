@@ -163,6 +163,11 @@ class Replicator:
             rs._thread = None
 
     def close(self):
+        """Stops this replicator's pull threads and closes its handles.
+        LIFETIME RULE: close downstream replicators BEFORE their upstreams —
+        a pull thread holds its upstream's db handle (the reference keeps
+        ReplicatedDB alive by shared_ptr until pullers drop,
+        rocksdb_replicator.cpp:145-151; C handles have no refcount)."""
         with self._mu:
             dbs = list(self._dbs.values())
             self._dbs.clear()
