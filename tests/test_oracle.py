@@ -308,3 +308,40 @@ def test_compressible_generator_compresses(lib):
         qlen = plib.gra_snappy_decompress(buf.raw[:clen], clen, dbuf, len(blob) + 16)
         assert qlen == len(blob) and dbuf.raw[:qlen] == blob
     assert total_c < used * 0.5, f"poor compression: {total_c}/{used}"
+
+
+def test_2pc_markers_consume_no_seq(lib):
+    """5.7.fb WAL markers (Noop, BeginPrepare, EndPrepare/Commit/Rollback
+    xid records) decode and consume no sequence numbers."""
+    rep = (PyBatch().noop().begin_prepare().put(b"k", b"v")
+           .commit_xid(b"xid12345").data())
+    seq, cnt, recs = oracle_ffi.decode(lib, rep)
+    assert cnt == 1
+    assert [r.type for r in recs] == [0x0D, 0x09, 0x01, 0x0B]
+    assert [r.consumes_seq for r in recs] == [0, 0, 1, 0]
+    st = oracle_ffi.Store(lib, 1)
+    assert st.apply(0, rep)
+    assert st.latest_seq(0) == 1
+    assert st.get(0, b"k") == b"v"
+
+
+def test_store_bytes_diagnostic(lib):
+    st = oracle_ffi.Store(lib, 2)
+    assert lib.orc_store_bytes(st.h) == 0
+    st.apply(0, PyBatch().put(b"k" * 16, b"v" * 100).data())
+    assert lib.orc_store_bytes(st.h) >= 116
+
+
+def test_snappy_empty_and_single_byte(lib):
+    import ctypes as CT
+    import rocksplicator_amd as ra
+    plib = ra.load()
+    for data in (b"", b"x"):
+        cap = 64
+        buf = CT.create_string_buffer(cap)
+        clen = plib.gra_snappy_compress(data, len(data), buf, cap)
+        assert clen > 0
+        out = CT.create_string_buffer(32)
+        qlen = plib.gra_snappy_decompress(buf.raw[:clen], clen, out, 32)
+        assert qlen == len(data)
+        assert out.raw[:qlen] == data
