@@ -428,3 +428,57 @@ def test_multi_shard_flush_churn_soak(olib):
     lrep.close()
     fe.close()
     le.close()
+
+
+def test_follower_restart_resumes_by_repull(olib):
+    """Recovery story: replication resume is inherent — a
+    restarted follower asks from LatestSequenceNumber()+1; a fresh (empty)
+    follower re-pulls everything and converges exactly."""
+    leader = ra.Engine(nshards=2, merge_op=1, retain_log=1)
+    ldb = leader.open(0)
+    one = (1).to_bytes(8, "little")
+    for i in range(300):
+        ldb.write_leader(PyBatch().merge(b"ctr", one).data())
+    f1 = ra.Engine(nshards=2, merge_op=1)
+    fdb1 = f1.open(0)
+    assert replicator.catch_up(ldb, fdb1, f1)
+    assert int.from_bytes(fdb1.get(b"ctr"), "little") == 300
+    f1.close()  # "crash" — the HBM store is gone with the engine
+    # more writes while the follower is down
+    for i in range(50):
+        ldb.write_leader(PyBatch().merge(b"ctr", one).data())
+    # restarted follower: empty store, seq 0 -> full re-pull from the log
+    f2 = ra.Engine(nshards=2, merge_op=1)
+    fdb2 = f2.open(0)
+    assert fdb2.latest_seq() == 0
+    assert replicator.catch_up(ldb, fdb2, f2)
+    assert fdb2.latest_seq() == 350
+    assert int.from_bytes(fdb2.get(b"ctr"), "little") == 350
+    f2.close()
+    leader.close()
+
+
+def test_wire_server_survives_malformed_frames(olib):
+    """A garbage client must not take the server down for other pullers."""
+    import socket
+    from rocksplicator_amd import wire
+    leader = ra.Engine(nshards=1, retain_log=1)
+    ldb = leader.open(0)
+    ldb.write_leader(PyBatch().put(b"k", b"v").data())
+    srv = wire.UpdateServer()
+    srv.register("db", ldb)
+    # garbage bytes on one connection
+    bad = socket.create_connection(("127.0.0.1", srv.port))
+    bad.sendall(b"\xde\xad\xbe\xef" * 8)
+    bad.close()
+    # unknown db name -> empty response, connection stays usable
+    r = wire.RemoteUpstream("127.0.0.1", srv.port, "nope")
+    assert r.get_updates(0) == []
+    r.close()
+    # a real puller still works
+    r2 = wire.RemoteUpstream("127.0.0.1", srv.port, "db")
+    ups = r2.get_updates(0)
+    assert len(ups) == 1 and ups[0][0] == 1
+    r2.close()
+    srv.close()
+    leader.close()
