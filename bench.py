@@ -385,7 +385,7 @@ def main():
                       "records": st.records},
     }
 
-    if rank == 0 and args.cpu_baseline:
+    if rank == 0 and world == 1 and args.cpu_baseline:  # contract: N=1 only
         if args.snappy:
             result["cpu_baseline"] = cpu_baseline_leg(
                 comp, comp_used.value, cdescs, n_upd, args.nshards,
