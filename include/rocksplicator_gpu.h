@@ -134,6 +134,15 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
                  const uint8_t *keybuf, size_t keybuf_len, uint8_t *valbuf,
                  uint32_t val_stride, GraGetResult *out);
 
+/* Full-store content checksum for any-size parity ("checksum of
+ * checksums"): per record FNV-1a over (seq LE8 | type | key_len LE4 |
+ * val_len LE4 | key bytes | val bytes), combined per shard by u64
+ * ADDITION — order/representation independent, so it can be compared
+ * against the oracle's orc_shard_checksum on identically applied streams
+ * at ANY size. Computed on-device over the run store (host-origin runs
+ * folded on the host). Call gra_flush first. */
+int gra_shard_checksum(GraDb *db, uint64_t *out);
+
 /* ---------------- leader update-serving ----------------
  * ≅ ReplicatedDB::handleReplicateRequest batch read-out (replicated_db.cpp:
  * 435-575): serve retained batches with base seq > since_seq, up to

@@ -142,6 +142,14 @@ double orc_cpu_snappy_apply_bench(OrcStore *s, const uint8_t *arena,
                                   const OrcUpdateDesc *descs, uint64_t n,
                                   int nthreads);
 
+/* Full-store content checksum for any-size parity ("checksum of
+ * checksums"): per record FNV-1a over (seq LE8 | type | key_len LE4 |
+ * val_len LE4 | key bytes | val bytes), combined per shard by unsigned
+ * 64-bit ADDITION (order-independent, so run/representation layout does
+ * not matter). Range tombstones hash with key=begin, val=end. Must match
+ * the engine's gra_shard_checksum on identically applied streams. */
+uint64_t orc_shard_checksum(const OrcStore *s, uint32_t shard);
+
 #ifdef __cplusplus
 }
 #endif

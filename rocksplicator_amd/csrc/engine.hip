@@ -533,6 +533,48 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   }
 }
 
+/* ---- full-store checksum (parity at any size) ----
+ * Same record hash as the oracle's orc_shard_checksum: FNV-1a over
+ * (seq LE8 | type | key_len LE4 | val_len LE4 | key | val), u64-ADD
+ * combined (order-independent). */
+__host__ __device__ inline uint64_t rec_hash_cs(uint64_t seq, uint8_t type,
+                                                uint32_t klen, uint32_t vlen,
+                                                const uint8_t *key,
+                                                const uint8_t *val) {
+  uint64_t h = 1469598103934665603ULL;
+#define GRA_FOLD(b) h = (h ^ (uint8_t)(b)) * 1099511628211ULL
+  for (int i = 0; i < 8; i++) GRA_FOLD(seq >> (8 * i));
+  GRA_FOLD(type);
+  for (int i = 0; i < 4; i++) GRA_FOLD(klen >> (8 * i));
+  for (int i = 0; i < 4; i++) GRA_FOLD(vlen >> (8 * i));
+  for (uint32_t i = 0; i < klen; i++) GRA_FOLD(key[i]);
+  for (uint32_t i = 0; i < vlen; i++) GRA_FOLD(val[i]);
+#undef GRA_FOLD
+  return h;
+}
+
+__global__ void k_checksum(const uint8_t *__restrict__ store,
+                           const RunView *__restrict__ runs, uint32_t nruns,
+                           unsigned long long *__restrict__ out) {
+  uint64_t local = 0;
+  for (uint32_t r = blockIdx.x; r < nruns; r += gridDim.x) {
+    RunView rv = runs[r];
+    const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+    const uint8_t *pay = store + rv.payload_off;
+    for (uint32_t i = threadIdx.x; i < rv.n_entries; i += blockDim.x) {
+      wb::RecHdr h = hdrs[i];
+      uint32_t rel = h.kv_off - rv.pay_rel_base;
+      local += rec_hash_cs(h.seq, h.type, h.key_len, h.val_len, pay + rel,
+                           pay + rel + h.key_len);
+    }
+  }
+  /* wave-reduce then one atomic per wave */
+  for (int ofs = 32; ofs > 0; ofs >>= 1)
+    local += __shfl_down(local, ofs, 64);
+  if ((threadIdx.x & 63) == 0 && local)
+    atomicAdd(out, (unsigned long long)local);
+}
+
 /* ---------------- host engine ---------------- */
 
 struct Stats {
@@ -1557,6 +1599,60 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
     rc = GRA_OK;
   } while (0);
   if (rc != GRA_OK) g_err = "gra_multiget: device op failed";
+  return rc;
+}
+
+int gra_shard_checksum(GraDb *db, uint64_t *out) {
+  GraEngine *e = db->e;
+  ShardState &ss = e->shards[db->shard];
+  std::vector<RunView> views;
+  uint64_t host_sum = 0;
+  {
+    std::lock_guard<std::mutex> lk(ss.mu);
+    for (const auto &rp : ss.runs) {
+      const Run &r = *rp;
+      if (r.n_entries == 0) continue;
+      if (r.hdr_cur == UINT64_MAX) { /* host-origin: fold on the host */
+        const wb::RecHdr *h = (const wb::RecHdr *)r.hdrs.data();
+        for (uint32_t i = 0; i < r.n_entries; i++)
+          host_sum += rec_hash_cs(h[i].seq, h[i].type, h[i].key_len,
+                                  h[i].val_len, r.payload.data() + h[i].kv_off,
+                                  r.payload.data() + h[i].kv_off + h[i].key_len);
+      } else {
+        views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
+      }
+    }
+  }
+  if (views.empty()) {
+    *out = host_sum;
+    return GRA_OK;
+  }
+  std::lock_guard<std::mutex> lk(e->mu);
+  RunView *d_runs = nullptr;
+  unsigned long long *d_sum = nullptr;
+  int rc = GRA_ERR;
+  do {
+    if (hipMalloc(&d_runs, views.size() * sizeof(RunView)) != hipSuccess ||
+        hipMalloc(&d_sum, 8) != hipSuccess)
+      break;
+    if (hipMemcpy(d_runs, views.data(), views.size() * sizeof(RunView),
+                  hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemset(d_sum, 0, 8) != hipSuccess)
+      break;
+    uint32_t nb = views.size() < 1024 ? (uint32_t)views.size() : 1024;
+    hipLaunchKernelGGL(k_checksum, dim3(nb), dim3(256), 0, e->stream, e->d_store,
+                       d_runs, (uint32_t)views.size(), d_sum);
+    if (hipGetLastError() != hipSuccess) break;
+    unsigned long long sum = 0;
+    if (hipStreamSynchronize(e->stream) != hipSuccess ||
+        hipMemcpy(&sum, d_sum, 8, hipMemcpyDeviceToHost) != hipSuccess)
+      break;
+    *out = host_sum + (uint64_t)sum;
+    rc = GRA_OK;
+  } while (0);
+  if (rc != GRA_OK) g_err = "gra_shard_checksum: device op failed";
+  if (d_runs) (void)hipFree(d_runs);
+  if (d_sum) (void)hipFree(d_sum);
   return rc;
 }
 

@@ -135,6 +135,7 @@ def load():
     lib.gra_multiget.argtypes = [C.c_void_p, C.c_uint32, C.POINTER(GraKeyRef),
                                  C.c_char_p, C.c_size_t, C.c_char_p,
                                  C.c_uint32, C.POINTER(GraGetResult)]
+    lib.gra_shard_checksum.argtypes = [C.c_void_p, C.POINTER(C.c_uint64)]
     lib.gra_pin_alloc.argtypes = [C.c_void_p, C.c_size_t, C.POINTER(C.POINTER(C.c_uint8))]
     lib.gra_pin_free.argtypes = [C.c_void_p, C.POINTER(C.c_uint8)]
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
@@ -374,6 +375,15 @@ class Db:
             else:  # needs host (merge fold / host-origin runs)
                 res.append(self.get(keys[i]))
         return res
+
+    def checksum(self):
+        """Order-independent full-store content checksum (device-computed;
+        comparable with the oracle's orc_shard_checksum at any size)."""
+        v = C.c_uint64()
+        rc = self.lib.gra_shard_checksum(self.h, C.byref(v))
+        if rc != GRA_OK:
+            raise RuntimeError(f"gra_shard_checksum rc={rc}: {last_error(self.lib)}")
+        return v.value
 
     def counters(self):
         """Per-db stats ≅ the reference's per-db counter fan-out

@@ -63,6 +63,8 @@ def load():
         C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t,
         C.c_char_p, C.c_size_t, C.POINTER(C.c_size_t),
     ]
+    lib.orc_shard_checksum.argtypes = [C.c_void_p, C.c_uint32]
+    lib.orc_shard_checksum.restype = C.c_uint64
     lib.orc_store_bytes.argtypes = [C.c_void_p]
     lib.orc_store_bytes.restype = C.c_uint64
     lib.orc_cpu_apply_bench.restype = C.c_double

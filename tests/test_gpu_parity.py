@@ -688,3 +688,36 @@ def test_tick_h2d_rejects_snappy_and_device_uploads(olib):
     rep.tick(0, 100)  # the supported path still works
     rep.sync()
     e.close()
+
+
+def test_full_store_checksum_parity(olib):
+    """'Checksum of checksums' (any-size parity property): the device store's
+    order-independent per-record content hash must equal the oracle's on the
+    same stream — full-store bit equality without per-key probing."""
+    nshards, nupd = 256, 200000
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=200, kind=2,
+                                       key_space=1 << 14, seed=424242)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd)
+    e = ra.Engine(nshards=nshards, store_bytes=2 << 30)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    for first in range(0, nupd, 51200):
+        rep.tick(first, min(51200, nupd - first))
+    rep.sync()
+    mismatches = []
+    for s in range(nshards):
+        db = e.open(s)
+        gpu = db.checksum()
+        orc = olib.orc_shard_checksum(ost.h, s)
+        if gpu != orc:
+            mismatches.append((s, hex(gpu), hex(orc)))
+        db.close()
+    assert not mismatches, mismatches[:5]
+    # mixed host(leader)+device runs also checksum consistently
+    db = e.open(0)
+    rep2 = PyBatch().put(b"hostside", b"entry").data()
+    db.write_leader(rep2)
+    assert ost.apply(0, rep2)
+    assert db.checksum() == olib.orc_shard_checksum(ost.h, 0)
+    e.close()
