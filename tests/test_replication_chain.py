@@ -51,6 +51,10 @@ def test_leader_follower_catchup(olib):
         assert fdbs[s].latest_seq() == ldbs[s].latest_seq() == ost.latest_seq(s)
         for k in keys:
             assert fdbs[s].get(k) == ost.get(s, k), (s, k)
+        # whole-store equality in one number: leader (host runs), follower
+        # (device runs) and oracle all checksum identically
+        assert (fdbs[s].checksum() == ldbs[s].checksum()
+                == olib.orc_shard_checksum(ost.h, s)), s
     leader.close()
     follower.close()
 
@@ -70,6 +74,8 @@ def test_chain_topology(olib):
     assert replicator.catch_up(ldb, mdb, mid)
     assert replicator.catch_up(mdb, tdb, tail)
     assert tdb.latest_seq() == mdb.latest_seq() == ldb.latest_seq() == 120
+    assert (tdb.checksum() == mdb.checksum() == ldb.checksum()
+            == olib.orc_shard_checksum(ost.h, 0))
     for i in range(17):
         k = f"k{i}".encode()
         assert tdb.get(k) == mdb.get(k) == ost.get(0, k)
