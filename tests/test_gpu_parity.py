@@ -721,3 +721,30 @@ def test_full_store_checksum_parity(olib):
     assert ost.apply(0, rep2)
     assert db.checksum() == olib.orc_shard_checksum(ost.h, 0)
     e.close()
+
+
+def test_tiny_staging_forces_overflow_ticks(olib):
+    """A deliberately tiny staging buffer forces the writer-overflow tick
+    path (chunk reservation fails -> abandon + tick + retry) on every few
+    updates; parity must hold."""
+    e = ra.Engine(nshards=2, staging_bytes=1 << 20)  # 1 MiB staging
+    dbs = [e.open(s) for s in range(2)]
+    ost = oracle_ffi.Store(olib, 2)
+    big = bytes(range(256)) * 128  # 32 KiB values
+    for i in range(60):  # ~2 MB of blobs through a 1 MB buffer
+        s = i & 1
+        rep = PyBatch().put(f"k{i:02d}".encode(), big).data()
+        assert dbs[s].handle_replicate_response(rep)
+        assert ost.apply(s, rep)
+    e.flush()
+    for s in range(2):
+        assert dbs[s].latest_seq() == ost.latest_seq(s)
+        assert dbs[s].checksum() == olib.orc_shard_checksum(ost.h, s)
+    st = e.stats()
+    assert st.ticks >= 2  # the overflow path actually forced multiple ticks
+    e.close()
+
+
+def test_bad_device_index_fails_gracefully(olib):
+    with pytest.raises(RuntimeError):
+        ra.Engine(nshards=1, device=99)
