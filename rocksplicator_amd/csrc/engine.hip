@@ -736,6 +736,9 @@ int GraEngine::init(const GraEngineOpts &o) {
   max_upd = 1u << 20;
   task_cap = 4u << 20;
   group_cap = opts.nshards + 1 > 65536 ? opts.nshards + 1 : 65536;
+  (void)hipGetLastError(); /* consume any sticky per-thread error a prior
+                              (possibly intentionally failing) call left —
+                              our launch checks would misattribute it */
   int ndev = 0;
   hipError_t de = hipGetDeviceCount(&ndev);
   if (de != hipSuccess || ndev == 0) {
@@ -743,7 +746,14 @@ int GraEngine::init(const GraEngineOpts &o) {
             "(no CPU fallback by design)";
     return GRA_NO_GPU;
   }
-  if (opts.device >= 0) HIP_TRY(hipSetDevice(opts.device));
+  if (opts.device >= 0) {
+    hipError_t se = hipSetDevice(opts.device);
+    if (se != hipSuccess) {
+      g_err = std::string("hipSetDevice: ") + hipGetErrorString(se);
+      (void)hipGetLastError(); /* clear the sticky error we just caused */
+      return GRA_ERR;
+    }
+  }
   HIP_TRY(hipStreamCreate(&stream));
   HIP_TRY(hipStreamCreate(&copyout));
   HIP_TRY(hipMalloc(&d_store, opts.store_bytes + 16));
