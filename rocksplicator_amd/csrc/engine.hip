@@ -1024,6 +1024,17 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
       const GroupDesc &gd = sl.h_groups[g];
       ShardState &ss = shards[rd.shard];
       std::lock_guard<std::mutex> lk(ss.mu);
+      if (rd.base_seq == 0) {
+        /* place overflow (store full in non-ring mode): NOTHING of this tick
+         * was applied — roll back to the durable boundary and poison; the
+         * condition persists until the operator gives the store room, like
+         * the reference's repeated DB::Write failures on a full disk
+         * (base_seq==0 is the discriminator: real seqs start at 1). */
+        ss.poisoned = true;
+        ss.next_seq = ss.durable_seq + 1;
+        ss.cnt_failures++;
+        continue;
+      }
       uint32_t keep_recs = 0;
       uint64_t keep_seq = rd.base_seq - 1;
       bool bad = ss.poisoned; /* shard already failed earlier this tick */

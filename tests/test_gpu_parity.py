@@ -748,3 +748,27 @@ def test_tiny_staging_forces_overflow_ticks(olib):
 def test_bad_device_index_fails_gracefully(olib):
     with pytest.raises(RuntimeError):
         ra.Engine(nshards=1, device=99)
+
+
+def test_store_full_rolls_back_and_poisons(olib):
+    """Non-ring store exhaustion (GRA_FULL condition): the overflowing
+    tick applies nothing; seqs roll back to the durable boundary and the
+    shard poisons until the operator intervenes — no silent seq advance."""
+    e = ra.Engine(nshards=1, store_bytes=1 << 20, store_ring=0)  # 1 MiB store
+    db = e.open(0)
+    small = PyBatch().put(b"first", b"x").data()
+    assert db.handle_replicate_response(small)
+    e.flush()
+    assert db.latest_seq() == 1
+    big_val = bytes(512 * 1024)  # two of these exceed the 1 MiB arena
+    for i in range(4):
+        db.handle_replicate_response(
+            PyBatch().put(f"big{i}".encode(), big_val).data())
+    e.flush()
+    c = db.counters()
+    assert c["apply_failures"] >= 1
+    assert db.latest_seq() <= 3  # durable boundary, not the optimistic 5
+    assert not db.handle_replicate_response(small)  # poisoned: fail once
+    # durable data intact
+    assert db.get(b"first") == b"x"
+    e.close()
