@@ -1033,6 +1033,10 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
         ss.poisoned = true;
         ss.next_seq = ss.durable_seq + 1;
         ss.cnt_failures++;
+        while (!ss.log.empty() && ss.log.back().base_seq > ss.durable_seq) {
+          ss.log_used -= ss.log.back().rep.size();
+          ss.log.pop_back();
+        }
         continue;
       }
       uint32_t keep_recs = 0;
@@ -1083,6 +1087,13 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
       ss.poisoned = true;
       ss.next_seq = ss.durable_seq + 1;
       ss.cnt_failures++; /* ≅ kReplicatorHandleResponseFailure */
+      /* drop retained-log entries past the rolled-back boundary: a batch
+       * that failed validation must never be re-served downstream (the
+       * reference's WAL only ever contains accepted batches) */
+      while (!ss.log.empty() && ss.log.back().base_seq > ss.durable_seq) {
+        ss.log_used -= ss.log.back().rep.size();
+        ss.log.pop_back();
+      }
     }
     for (int i = 0; i < kEventsPerTick; i++) put_event(t.ev[i]);
     sl.busy = false;

@@ -490,3 +490,31 @@ def test_wire_server_survives_malformed_frames(olib):
     r2.close()
     srv.close()
     leader.close()
+
+
+def test_rejected_batches_never_served_downstream(olib):
+    """A batch that fails GPU validation after being optimistically retained
+    must be pruned from the serving log (the reference's WAL only ever holds
+    accepted batches) — a chained follower pulling afterwards sees only the
+    durable prefix."""
+    mid = ra.Engine(nshards=1, retain_log=1)
+    mdb = mid.open(0)
+    good1 = PyBatch().put(b"a", b"1").data()
+    bad = bytearray(PyBatch().put(b"x", b"y").data())
+    bad[8] = 9  # count mismatch
+    good2 = PyBatch().put(b"b", b"2").data()
+    assert mdb.handle_replicate_response(good1)
+    assert mdb.handle_replicate_response(bytes(bad))
+    assert mdb.handle_replicate_response(good2)  # rolled back with the bad one
+    mid.flush()
+    assert mdb.latest_seq() == 1
+    ups = mdb.get_updates(0)
+    assert [u[0] for u in ups] == [1]  # only the durable batch is served
+    assert ups[0][2] == good1
+    # recovery: re-apply after the fail-once signal; serving log follows
+    assert not mdb.handle_replicate_response(good2)
+    assert mdb.handle_replicate_response(good2)
+    mid.flush()
+    ups = mdb.get_updates(0)
+    assert [u[0] for u in ups] == [1, 2]
+    mid.close()
