@@ -798,6 +798,8 @@ int GraEngine::init(const GraEngineOpts &o) {
 
 GraEngine::~GraEngine() {
   (void)hipStreamSynchronize(stream);
+  (void)hipStreamSynchronize(copyout); /* pending rundesc/ok D2H target the
+                                          pinned slot buffers freed below */
   for (auto &t : pending)
     for (int i = 0; i < kEventsPerTick; i++)
       if (t.ev[i]) (void)hipEventDestroy(t.ev[i]);
