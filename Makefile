@@ -30,6 +30,21 @@ rocksplicator_amd/libgra.so: $(OBJS)
 oracle/libwb_oracle.so: oracle/wb_oracle.c oracle/wb_oracle.h
 	$(MAKE) -C oracle
 
+# auxiliary binaries (C++ chain test, streaming bench, kernel microbenches)
+tools: build/test_cpp_chain build/bench_stream build/micro_copy build/micro_snappy
+
+build/test_cpp_chain: scripts/test_cpp_chain.cpp include/rocksplicator_replicator.hpp include/rocksplicator_gpu.h rocksplicator_amd/libgra.so | build
+	$(HIPCC) $(HIPFLAGS) $< -Iinclude -Lrocksplicator_amd -lgra -Wl,-rpath,'$$ORIGIN/../rocksplicator_amd' -o $@
+
+build/bench_stream: scripts/bench_stream.cpp include/rocksplicator_gpu.h rocksplicator_amd/libgra.so | build
+	$(HIPCC) $(HIPFLAGS) $< -Lrocksplicator_amd -lgra -Wl,-rpath,'$$ORIGIN/../rocksplicator_amd' -o $@
+
+build/micro_copy: scripts/micro_copy.hip | build
+	$(HIPCC) $(HIPFLAGS) -x hip $< -o $@
+
+build/micro_snappy: scripts/micro_snappy.hip | build
+	$(HIPCC) $(HIPFLAGS) -x hip $< -o $@
+
 clean:
 	rm -rf build rocksplicator_amd/libgra.so
 	$(MAKE) -C oracle clean
