@@ -575,6 +575,14 @@ __global__ void k_checksum(const uint8_t *__restrict__ store,
     atomicAdd(out, (unsigned long long)local);
 }
 
+static bool gra_check_staging() {
+  static const bool on = [] {
+    const char *v = getenv("GRA_CHECK_STAGING");
+    return v && v[0] == '1';
+  }();
+  return on;
+}
+
 /* ---------------- host engine ---------------- */
 
 struct Stats {
@@ -1165,10 +1173,16 @@ int GraEngine::stream_tick_locked() {
                     without this, stale chunks write into recycled slots —
                     lost updates under flush churn) */
   next->closed.store(false, std::memory_order_release);
-  /* swap: new writers land in `next`; then quiesce `old` */
-  old->closed.store(true, std::memory_order_release);
+  /* swap: new writers land in `next`; then quiesce `old`.
+   * The close/quiesce pair is a Dekker handshake with the writers'
+   * (writers++ ; closed.load) pair: with a plain release store the
+   * builder's closed=true can reorder after its writers read (x86
+   * StoreLoad), letting quiesce pass while a writer that never saw
+   * `closed` stages one more update — observed as rare single-update
+   * loss under flush churn. seq_cst on both sides closes it. */
+  old->closed.store(true, std::memory_order_seq_cst);
   cur_stage.store(next, std::memory_order_release);
-  while (old->writers.load(std::memory_order_acquire) != 0)
+  while (old->writers.load(std::memory_order_seq_cst) != 0)
     std::this_thread::yield();
   uint32_t nall = old->nslots.load(std::memory_order_relaxed);
   if (nall > max_upd) nall = max_upd;
@@ -1180,8 +1194,9 @@ int GraEngine::stream_tick_locked() {
     if (old->descs[i].len) cnt[old->descs[i].shard + 1]++;
   for (uint32_t s = 0; s < opts.nshards; s++) cnt[s + 1] += cnt[s];
   uint32_t n = cnt[opts.nshards];
-  { /* invariant: every successful HandleReplicateResponse of this buffer
-       generation must surface exactly once in the tick build */
+  if (gra_check_staging()) {
+    /* invariant: every successful HandleReplicateResponse of this buffer
+     * generation must surface exactly once in the tick build */
     uint32_t expect = old->staged.load(std::memory_order_relaxed);
     if (n != expect)
       fprintf(stderr,
@@ -1330,8 +1345,9 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
   bool staged = false;
   for (int attempt = 0; attempt < 100000 && !staged; attempt++) {
     GraEngine::StageBuf *sb = e->cur_stage.load(std::memory_order_acquire);
-    sb->writers.fetch_add(1, std::memory_order_acq_rel);
-    if (sb->closed.load(std::memory_order_acquire)) {
+    sb->writers.fetch_add(1, std::memory_order_seq_cst);
+    if (sb->closed.load(std::memory_order_seq_cst)) { /* Dekker pair with the
+                                                         builder's close+quiesce */
       sb->writers.fetch_sub(1, std::memory_order_acq_rel);
       std::this_thread::yield();
       continue;
@@ -1372,7 +1388,9 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     ck.used += len;
     ck.slot++;
     ck.slots_left--;
-    sb->staged.fetch_add(1, std::memory_order_relaxed);
+    if (gra_check_staging()) /* invariant counter is a hot shared RMW:
+                                debug builds/soaks only (GRA_CHECK_STAGING=1) */
+      sb->staged.fetch_add(1, std::memory_order_relaxed);
     sb->writers.fetch_sub(1, std::memory_order_release);
     staged = true;
   }
@@ -1382,18 +1400,21 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
     return 0;
   }
   /* success: bookkeeping + retention (AFTER staging so a failure above
-   * never leaves a retained-log entry for an un-applied batch) */
-  {
+   * never leaves a retained-log entry for an un-applied batch); counters
+   * are relaxed atomics — no lock on the hot path */
+  if (e->opts.retain_log) {
     std::lock_guard<std::mutex> lk(ss.mu);
-    if (e->opts.retain_log) retain_locked(e, ss, base, count, ts, rep, len);
-    ss.cnt_updates++;
-    ss.cnt_in_bytes += len; /* ≅ kReplicatorInBytes (replicated_db.cpp:409) */
-    if (ts != 0) { /* ≅ kReplicatorLatency (replicated_db.cpp:370-374) */
-      int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
-          std::chrono::system_clock::now().time_since_epoch()).count());
-      ss.lat_sum_ms += (uint64_t)(now_ms > ts ? now_ms - ts : 0);
-      ss.lat_n++;
-    }
+    retain_locked(e, ss, base, count, ts, rep, len);
+  }
+  ss.cnt_updates.fetch_add(1, std::memory_order_relaxed);
+  ss.cnt_in_bytes.fetch_add(len, std::memory_order_relaxed);
+  /* ≅ kReplicatorInBytes (replicated_db.cpp:409) */
+  if (ts != 0) { /* ≅ kReplicatorLatency (replicated_db.cpp:370-374) */
+    int64_t now_ms = (int64_t)(std::chrono::duration_cast<std::chrono::milliseconds>(
+        std::chrono::system_clock::now().time_since_epoch()).count());
+    ss.lat_sum_ms.fetch_add((uint64_t)(now_ms > ts ? now_ms - ts : 0),
+                            std::memory_order_relaxed);
+    ss.lat_n.fetch_add(1, std::memory_order_relaxed);
   }
   return 1;
 }
@@ -1482,13 +1503,13 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
 int gra_db_counters(GraDb *db, GraDbCounters *out) {
   ShardState &ss = db->e->shards[db->shard];
   std::lock_guard<std::mutex> lk(ss.mu);
-  out->updates_applied = ss.cnt_updates;
-  out->in_bytes = ss.cnt_in_bytes;
-  out->apply_failures = ss.cnt_failures;
-  out->updates_served = ss.cnt_served;
-  out->out_bytes = ss.cnt_out_bytes;
-  out->latency_ms_sum = ss.lat_sum_ms;
-  out->latency_samples = ss.lat_n;
+  out->updates_applied = ss.cnt_updates.load(std::memory_order_relaxed);
+  out->in_bytes = ss.cnt_in_bytes.load(std::memory_order_relaxed);
+  out->apply_failures = ss.cnt_failures.load(std::memory_order_relaxed);
+  out->updates_served = ss.cnt_served.load(std::memory_order_relaxed);
+  out->out_bytes = ss.cnt_out_bytes.load(std::memory_order_relaxed);
+  out->latency_ms_sum = ss.lat_sum_ms.load(std::memory_order_relaxed);
+  out->latency_samples = ss.lat_n.load(std::memory_order_relaxed);
   out->latest_seq = ss.poisoned ? ss.durable_seq : ss.next_seq - 1;
   return GRA_OK;
 }

@@ -13,6 +13,7 @@
  * its seq.
  */
 #pragma once
+#include <atomic>
 #include <condition_variable>
 #include <cstdint>
 #include <cstring>
@@ -64,10 +65,11 @@ struct ShardState {
   uint64_t acked_sent = 0, acked_confirmed = 0;
   std::condition_variable ack_cv;
   /* per-db counters ≅ replicator_stats.cpp:33-102's per-db fan-out
-   * (replicator_in_bytes / _out_bytes / latency / failure counters) */
-  uint64_t cnt_updates = 0, cnt_in_bytes = 0, cnt_failures = 0;
-  uint64_t cnt_served = 0, cnt_out_bytes = 0;
-  uint64_t lat_sum_ms = 0, lat_n = 0;
+   * (replicator_in_bytes / _out_bytes / latency / failure counters);
+   * relaxed atomics so the hot ingest path never takes a lock for them */
+  std::atomic<uint64_t> cnt_updates{0}, cnt_in_bytes{0}, cnt_failures{0};
+  std::atomic<uint64_t> cnt_served{0}, cnt_out_bytes{0};
+  std::atomic<uint64_t> lat_sum_ms{0}, lat_n{0};
 };
 
 /* Get over a run list (newest last). merge_op: 0 concat, 1 u64add.

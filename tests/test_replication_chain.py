@@ -395,11 +395,13 @@ def test_tcp_two_followers_one_leader(olib):
         e.close()
 
 
-def test_multi_shard_flush_churn_soak(olib):
+def test_multi_shard_flush_churn_soak(olib, monkeypatch):
     """8 writer threads + 8 pull threads flushing concurrently: heavy
     staging-buffer swap churn. Caught a real bug once (a stale thread-local
     staging chunk surviving a buffer-generation swap silently dropped ~25%
     of updates while seqs converged) — keep it in the suite."""
+    import os
+    os.environ["GRA_CHECK_STAGING"] = "1"  # enable the staging invariant
     import threading
     le = ra.Engine(nshards=8, merge_op=1, retain_log=1)
     fe = ra.Engine(nshards=8, merge_op=1)
