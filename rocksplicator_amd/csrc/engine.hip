@@ -276,9 +276,9 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = wb::base_tag(r.tag);
     h.flags = cf4 ? 1 : 0;
-    /* stored key = [cf?4B][key bytes]; prefix covers the stored form */
-    h.kpref = cf4 ? r.cf_id
-                  : wb::key_prefix4(blobs + d.off + r.key_off, r.key_len);
+    h.kpref = 0; /* reserved; a key-prefix filter here measured NEGATIVE:
+                    emit would re-touch blob key bytes (+35us/tick cold)
+                    for no multiget win */
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -446,7 +446,6 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   if (q >= nq) return;
   const uint8_t *key = keybuf + keys[q].off;
   uint32_t klen = keys[q].len;
-  uint32_t qpref = wb::key_prefix4(key, klen);
   uint64_t term_seq = 0, term_ref = ~0ULL, merge_seq = 0, rd_seq = 0;
   for (uint32_t r = 0; r < nruns; r++) {
     RunView rv = runs[r];
@@ -465,8 +464,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
         if (h.seq > rd_seq) rd_seq = h.seq;
         continue;
       }
-      if (h.key_len != klen || h.kpref != qpref ||
-          dev_memcmp(pay + rel, key, klen) != 0)
+      if (h.key_len != klen || dev_memcmp(pay + rel, key, klen) != 0)
         continue;
       if (h.type == wb::kMerge) {
         if (h.seq > merge_seq) merge_seq = h.seq;

@@ -119,7 +119,7 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
-      h.kpref = cf4 ? rc.cf_id : wb::key_prefix4(rep + rc.key_off, rc.key_len);
+      h.kpref = 0; /* reserved (see k_emit) */
       hd[i] = h;
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);
