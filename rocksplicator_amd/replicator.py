@@ -5,6 +5,7 @@ LatestSequenceNumber and applies each via HandleReplicateResponse, retrying
 from the durable seq on failure. Used by the replication-chain tests and
 as the wiring template for a transport layer.
 """
+import threading
 
 
 def pull_once(upstream_db, follower_db, max_updates=50, observer=False):
@@ -34,8 +35,6 @@ def catch_up(upstream_db, follower_db, follower_engine, max_rounds=10000):
         rounds += 1
     return False
 
-
-import threading
 
 LEADER = "LEADER"
 FOLLOWER = "FOLLOWER"
