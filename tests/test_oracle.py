@@ -345,3 +345,27 @@ def test_snappy_empty_and_single_byte(lib):
         qlen = plib.gra_snappy_decompress(buf.raw[:clen], clen, out, 32)
         assert qlen == len(data)
         assert out.raw[:qlen] == data
+
+
+def test_varint_boundary_lengths(lib):
+    """Slice lengths at every varint32 width boundary round-trip and decode
+    with exact offsets (1/2/3-byte varints; 127/128, 16383/16384)."""
+    for n in (0, 1, 127, 128, 129, 16383, 16384, 16385, 70000):
+        key = bytes(min(n, 300)) or b"k"
+        val = b"\xAB" * n
+        rep = PyBatch().put(key, val).data()
+        seq, cnt, recs = oracle_ffi.decode(lib, rep)
+        assert cnt == 1
+        r = recs[0]
+        assert r.val_len == n
+        assert rep[r.val_off:r.val_off + r.val_len] == val
+        # oracle == product builder bytes
+        import rocksplicator_amd as ra
+        assert ra.Batch().put(key, val).data() == rep
+
+
+def test_header_seq_extremes(lib):
+    for seq in (0, 1, 2**32 - 1, 2**32, 2**63, 2**64 - 1):
+        rep = PyBatch(seq=seq).put(b"k", b"v").data()
+        got_seq, cnt, recs = oracle_ffi.decode(lib, rep)
+        assert got_seq == seq and recs[0].seq == seq
