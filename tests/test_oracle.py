@@ -369,3 +369,78 @@ def test_header_seq_extremes(lib):
         rep = PyBatch(seq=seq).put(b"k", b"v").data()
         got_seq, cnt, recs = oracle_ffi.decode(lib, rep)
         assert got_seq == seq and recs[0].seq == seq
+
+def test_snappy_rare_elements(lib):
+    """Format corners real encoders rarely emit but decoders must accept:
+    4-byte-offset copies (tag 11) and 2/3-byte literal lengths (tags 61/62)."""
+    import ctypes as CT
+    import rocksplicator_amd as ra
+    plib = ra.load()
+    lib.orc_snappy_decompress.argtypes = [
+        CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t,
+        CT.POINTER(CT.c_size_t)]
+
+    def both(v, want):
+        dbuf = CT.create_string_buffer(len(want) + 16)
+        dlen = CT.c_size_t()
+        assert lib.orc_snappy_decompress(v, len(v), dbuf, len(want) + 16,
+                                         CT.byref(dlen)) == 0
+        assert dbuf.raw[:dlen.value] == want
+        qbuf = CT.create_string_buffer(len(want) + 16)
+        assert plib.gra_snappy_decompress(v, len(v), qbuf,
+                                          len(want) + 16) == len(want)
+        assert qbuf.raw[:len(want)] == want
+
+    # 4-byte-offset copy: "abcd" + copy(off=4 as LE32, len=8)
+    v = bytes([0x0C, 0x0C]) + b"abcd" + bytes([((8 - 1) << 2) | 3, 4, 0, 0, 0])
+    both(v, b"abcd" * 3)
+    # 2-byte literal length (tag 61): 300-byte literal
+    lit = bytes(range(256)) + bytes(44)
+    n = len(lit) - 1
+    v = bytes([0xAC, 0x02, 61 << 2, n & 0xFF, n >> 8]) + lit  # varint(300)=AC 02
+    both(v, lit)
+    # 3-byte literal length (tag 62): 70000-byte literal
+    lit = (bytes(range(256)) * 274)[:70000]
+    n = len(lit) - 1
+    v = bytes([0xF0, 0xA2, 0x04, 62 << 2, n & 0xFF, (n >> 8) & 0xFF, n >> 16]) + lit
+    both(v, lit)
+
+
+def test_snappy_fuzz_crossimpl(lib):
+    """Seeded fuzz: mixed-entropy buffers through both codecs in both
+    directions (oracle<->product), 40 cases."""
+    import ctypes as CT
+    import rocksplicator_amd as ra
+    plib = ra.load()
+    lib.orc_snappy_compress.restype = CT.c_size_t
+    lib.orc_snappy_compress.argtypes = [CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t]
+    lib.orc_snappy_decompress.argtypes = [
+        CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t,
+        CT.POINTER(CT.c_size_t)]
+    rng = random.Random(0xC0DEC)
+    for case in range(40):
+        n = rng.randrange(0, 9000)
+        parts, left = [], n
+        while left > 0:
+            m = min(left, rng.randrange(1, 400))
+            if rng.random() < 0.5:
+                parts.append(rng.randbytes(rng.randrange(1, 9)) * (m // 8 + 1))
+                parts[-1] = parts[-1][:m]
+            else:
+                parts.append(rng.randbytes(m))
+            left -= m
+        data = b"".join(parts)
+        cap = len(data) + len(data) // 6 + 64
+        obuf = CT.create_string_buffer(cap)
+        clen = lib.orc_snappy_compress(data, len(data), obuf, cap)
+        pbuf = CT.create_string_buffer(cap)
+        pclen = plib.gra_snappy_compress(data, len(data), pbuf, cap)
+        dbuf = CT.create_string_buffer(len(data) + 16)
+        dlen = CT.c_size_t()
+        assert lib.orc_snappy_decompress(pbuf.raw[:pclen], pclen, dbuf,
+                                         len(data) + 16, CT.byref(dlen)) == 0
+        assert dbuf.raw[:dlen.value] == data, case
+        qbuf = CT.create_string_buffer(len(data) + 16)
+        assert plib.gra_snappy_decompress(obuf.raw[:clen], clen, qbuf,
+                                          len(data) + 16) == len(data)
+        assert qbuf.raw[:len(data)] == data, case
