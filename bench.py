@@ -332,8 +332,10 @@ def main():
     if os.path.exists(args.traffic_file):
         try:
             tf = json.load(open(args.traffic_file))
-            if (tf.get("workload", {}).get("val_len") == args.val_len
-                    and tf.get("workload", {}).get("nshards") == args.nshards):
+            wk = tf.get("workload", {})
+            if (wk.get("val_len") == args.val_len
+                    and wk.get("nshards") == args.nshards
+                    and wk.get("tick_updates") in (None, args.tick_updates)):
                 traffic = tf.get("hbm_bytes_per_copy_launch")
         except Exception:
             pass
