@@ -1733,13 +1733,18 @@ static int upload_common(GraEngine *e, GraReplay *r, const uint8_t *arena,
   }
   r->descs.resize(n);
   r->counts.resize(n);
+  /* validate everything BEFORE touching shard seq state: a failed upload
+   * must leave next_seq untouched for the caller to retry */
   for (uint64_t i = 0; i < n; i++) {
     const GraUpdateDesc &d = descs[i];
-    if (d.shard >= e->opts.nshards || d.off + d.len > arena_bytes ||
-        d.len < wb::kHeaderBytes) {
+    if (d.shard >= e->opts.nshards || d.len > arena_bytes ||
+        d.off > arena_bytes - d.len || d.len < wb::kHeaderBytes) {
       g_err = "gra_upload: bad desc";
       return GRA_ERR;
     }
+  }
+  for (uint64_t i = 0; i < n; i++) {
+    const GraUpdateDesc &d = descs[i];
     uint32_t count = host_counts ? host_counts[i]
                                  : wb::fixed32_le(arena + d.off + 8);
     ShardState &ss = e->shards[d.shard];
@@ -1822,6 +1827,11 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
   std::vector<SnapTask> tasks(n);
   uint64_t scratch = 0;
   for (uint64_t i = 0; i < n; i++) {
+    if (descs[i].len > comp_bytes || descs[i].off > comp_bytes - descs[i].len) {
+      g_err = "gra_upload_snappy: comp desc out of range";
+      gra_replay_destroy(r);
+      return GRA_ERR;
+    }
     tasks[i] = {descs[i].off, descs[i].len, ulens[i]};
     udescs[i].shard = descs[i].shard;
     udescs[i].len = ulens[i];

@@ -772,3 +772,44 @@ def test_store_full_rolls_back_and_poisons(olib):
     # durable data intact
     assert db.get(b"first") == b"x"
     e.close()
+
+
+@pytest.mark.gpu
+def test_failed_upload_leaves_seq_state_untouched(olib):
+    """A rejected upload (bad descriptor) must not advance any shard's
+    next_seq: the engine validates every descriptor before assigning seqs,
+    so a caller can fix the input and retry without a seq gap."""
+    import ctypes as C
+    eng = ra.Engine(nshards=4)
+    try:
+        db = eng.open(1)
+        # seed one good batch so the shard has nonzero seq state
+        assert db.handle_replicate_response(ra.Batch().put(b"k", b"v").data(), 1)
+        eng.flush()
+        seq0 = db.latest_seq()
+
+        arena, used, descs = ra.ffi.gen_stream(
+            nshards=4, n_updates=64, key_len=16, val_len=64, kind=0, seed=42)
+        descs[50].off = used + 1000  # out of range -> whole upload rejected
+        with pytest.raises(RuntimeError, match="bad desc"):
+            eng.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, 64)
+        assert db.latest_seq() == seq0
+
+        # fixed input applies with contiguous seqs (no gap from the failure)
+        arena2, used2, descs2 = ra.ffi.gen_stream(
+            nshards=4, n_updates=64, key_len=16, val_len=64, kind=0, seed=42)
+        rep = eng.upload(C.cast(arena2, C.POINTER(C.c_uint8)), used2, descs2, 64)
+        rep.tick(0, 64)
+        rep.sync()
+        n_shard1 = sum(1 for d in descs2 if d.shard == 1)
+        assert db.latest_seq() == seq0 + n_shard1
+
+        # snappy comp-bounds validation rejects out-of-range comp descs
+        cdescs = (ra.ffi.GraUpdateDesc * 1)()
+        cdescs[0].shard, cdescs[0].len, cdescs[0].off = 0, 100, 64
+        buf = (C.c_uint8 * 128)()
+        with pytest.raises(RuntimeError, match="out of range"):
+            eng.upload_snappy(C.cast(buf, C.POINTER(C.c_uint8)), 128,
+                              cdescs, 1, [100])
+    finally:
+        eng.close()
