@@ -109,3 +109,75 @@ def test_cf_variants_decode(ops):
             assert r.cf_id == cf
             assert rep[r.key_off:r.key_off + r.key_len] == k
             assert rep[r.val_off:r.val_off + r.val_len] == want_val
+
+
+@settings(max_examples=200, deadline=None, derandomize=True)
+@given(data=st.data())
+def test_corrupt_reps_rejected_cleanly(data):
+    """Mutation fuzz: random byte flips/truncations of a valid rep must
+    never crash the oracle decoder/applier, and a rejected batch must not
+    advance the store's seq (all-or-nothing apply, two-pass validation)."""
+    lib = oracle_ffi.load()
+    ops = data.draw(st.lists(op, min_size=1, max_size=6))
+    b = PyBatch(seq=0)
+    for kind, a, v in ops:
+        if kind in ("put", "merge"):
+            getattr(b, kind)(a, v)
+        elif kind == "delete_range":
+            b.delete_range(a, v)
+        elif kind == "log_data":
+            b.log_data(a)
+        else:
+            getattr(b, kind)(a)
+    rep = bytearray(b.data())
+    mode = data.draw(st.sampled_from(["flip", "truncate", "extend"]))
+    if mode == "flip":
+        i = data.draw(st.integers(0, len(rep) - 1))
+        rep[i] ^= data.draw(st.integers(1, 255))
+    elif mode == "truncate":
+        rep = rep[:data.draw(st.integers(0, len(rep) - 1))]
+    else:
+        rep += bytes(data.draw(st.integers(1, 8)))
+    rep = bytes(rep)
+
+    try:
+        seq, cnt, recs = oracle_ffi.decode(lib, rep)
+        decoded_ok = True
+    except ValueError:
+        decoded_ok = False
+
+    st_ = oracle_ffi.Store(lib, 1)
+    before = st_.latest_seq(0)
+    applied = st_.apply(0, rep)
+    after = st_.latest_seq(0)
+    if applied:
+        assert decoded_ok  # apply is gated on the same two-pass decode
+        consumed = sum(1 for r in recs if r.consumes_seq)
+        assert after == before + consumed
+    else:
+        assert after == before  # rejected batch applies nothing
+
+
+@settings(max_examples=200, deadline=None, derandomize=True)
+@given(blob=st.binary(max_size=600))
+def test_snappy_decoder_rejects_garbage(blob):
+    """Arbitrary bytes into both snappy decoders: bounded, no crash, and
+    the two independent restatements agree on accept/reject AND on the
+    decoded bytes when they accept."""
+    import ctypes as CT
+    import rocksplicator_amd as ra
+    lib = oracle_ffi.load()
+    plib = ra.load()
+    lib.orc_snappy_decompress.argtypes = [
+        CT.c_char_p, CT.c_size_t, CT.c_char_p, CT.c_size_t,
+        CT.POINTER(CT.c_size_t)]
+    cap = 700
+    obuf = CT.create_string_buffer(cap + 16)
+    olen = CT.c_size_t()
+    orc = lib.orc_snappy_decompress(blob, len(blob), obuf, cap, CT.byref(olen))
+    pbuf = CT.create_string_buffer(cap + 16)
+    plen = plib.gra_snappy_decompress(blob, len(blob), pbuf, cap)
+    assert (orc == 0) == (plen != 0xFFFFFFFF), (orc, plen)
+    if orc == 0:
+        assert olen.value == plen
+        assert obuf.raw[:olen.value] == pbuf.raw[:plen]
