@@ -8,6 +8,7 @@
  */
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -388,5 +389,34 @@ int main(int argc, char **argv) {
     hipLaunchKernelGGL(k_v4<508>, dim3((n + 255) / 256), dim3(256), 0, 0,
                        d_comp, d_tasks, n, d_out);
   });
+
+  /* v2 on tasks SORTED by compressed length: lanes of a wave get
+   * similar-size streams, so nobody idles waiting for a long tail
+   * (candidate launch-order fix for wave-tail divergence; output
+   * offsets ride with the tasks, results land identically). */
+  {
+    std::vector<Task> sorted = tasks;
+    std::stable_sort(sorted.begin(), sorted.end(),
+                     [](const Task &a, const Task &b) {
+                       return a.comp_len < b.comp_len;
+                     });
+    Task *d_sorted;
+    CHECK(hipMalloc(&d_sorted, n * sizeof(Task)));
+    CHECK(hipMemcpy(d_sorted, sorted.data(), n * sizeof(Task),
+                    hipMemcpyHostToDevice));
+    run("v2-lensorted", [&] {
+      hipLaunchKernelGGL(k_v2<508>, dim3((n + 255) / 256), dim3(256), 0, 0,
+                         d_comp, d_sorted, n, d_out);
+    });
+    run("v2-lds512", [&] {
+      hipLaunchKernelGGL(k_v2<508>, dim3((n + 255) / 256), dim3(256), 0, 0,
+                         d_comp, d_tasks, n, d_out);
+    });
+    run("v2-lensorted", [&] {
+      hipLaunchKernelGGL(k_v2<508>, dim3((n + 255) / 256), dim3(256), 0, 0,
+                         d_comp, d_sorted, n, d_out);
+    });
+    CHECK(hipFree(d_sorted));
+  }
   return 0;
 }
