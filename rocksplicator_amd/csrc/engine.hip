@@ -84,6 +84,8 @@ struct CopyTask {
 };
 struct SnapTask { /* one compressed Update payload (config #5) */
   uint64_t comp_off;
+  uint64_t out_off; /* scratch-arena slot (rides with the task so the launch
+                       order can be length-sorted independently of descs) */
   uint32_t comp_len;
   uint32_t ulen;
 };
@@ -118,14 +120,13 @@ constexpr uint32_t kSnapStage = 508; /* 256 x (508+4) = 128 KiB LDS */
 
 __global__ void __launch_bounds__(256) k_snappy(
     const uint8_t *__restrict__ comp, const SnapTask *__restrict__ tasks,
-    const UpdDesc *__restrict__ descs, uint32_t n,
-    uint8_t *__restrict__ scratch, uint32_t *__restrict__ err_ring,
-    uint32_t tick) {
+    uint32_t n, uint8_t *__restrict__ scratch,
+    uint32_t *__restrict__ err_ring, uint32_t tick) {
   __shared__ uint8_t lds[256 * (kSnapStage + 4)];
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   SnapTask t = tasks[i];
-  uint8_t *dst = scratch + descs[i].off;
+  uint8_t *dst = scratch + t.out_off;
   uint32_t r;
   if (t.comp_len <= kSnapStage) {
     uint8_t *mine = lds + threadIdx.x * (kSnapStage + 4);
@@ -703,6 +704,11 @@ struct TickPlan {
   std::vector<GroupDesc> groups;
   uint64_t blob_bytes = 0;
   GroupDesc *d_groups = nullptr; /* device-cached copy (freed with replay) */
+  SnapTask *d_snap = nullptr;    /* window's snap tasks, LENGTH-SORTED so a
+                                    wave's lanes get similar-size streams
+                                    (+15% k_snappy, scripts/micro_snappy.hip
+                                    v2-lensorted; out_off rides with the
+                                    task, results land identically) */
 };
 
 struct GraReplay {
@@ -718,6 +724,7 @@ struct GraReplay {
    * scratch arena, blobs decompressed per tick by k_snappy */
   uint8_t *d_comp = nullptr;
   SnapTask *d_snaptasks = nullptr;
+  std::vector<SnapTask> snap_tasks; /* host copy for per-window sorting */
   bool snappy = false;
   std::map<std::pair<uint64_t, uint64_t>, TickPlan> plans; /* window cache */
 };
@@ -918,8 +925,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   uint32_t nb = (n + 255) / 256;
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena */
     hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, stream, d_comp,
-                       d_snaptasks, d_descw, n, (uint8_t *)d_blobs, d_err_ring,
-                       tick);
+                       d_snaptasks, n, (uint8_t *)d_blobs, d_err_ring, tick);
     HIP_TRY(hipGetLastError());
     HIP_TRY(rec(8)); /* after snappy */
   }
@@ -1832,7 +1838,7 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
       gra_replay_destroy(r);
       return GRA_ERR;
     }
-    tasks[i] = {descs[i].off, descs[i].len, ulens[i]};
+    tasks[i] = {descs[i].off, scratch, descs[i].len, ulens[i]};
     udescs[i].shard = descs[i].shard;
     udescs[i].len = ulens[i];
     udescs[i].off = scratch;
@@ -1860,6 +1866,7 @@ int gra_upload_snappy(GraEngine *e, const uint8_t *comp_arena,
     gra_replay_destroy(r);
     return rc;
   }
+  r->snap_tasks = std::move(tasks);
   *out = r;
   return GRA_OK;
 }
@@ -1870,8 +1877,10 @@ void gra_replay_destroy(GraReplay *r) {
   if (r->d_comp) (void)hipFree(r->d_comp);
   if (r->d_snaptasks) (void)hipFree(r->d_snaptasks);
   if (r->d_descs) (void)hipFree(r->d_descs);
-  for (auto &kv : r->plans)
+  for (auto &kv : r->plans) {
     if (kv.second.d_groups) (void)hipFree(kv.second.d_groups);
+    if (kv.second.d_snap) (void)hipFree(kv.second.d_snap);
+  }
   delete r;
 }
 
@@ -1905,6 +1914,23 @@ static TickPlan &plan_for(GraReplay *r, uint64_t first, uint64_t n) {
   } else {
     plan.d_groups = nullptr; /* fall back to per-tick upload */
   }
+  if (r->snappy && first + n <= r->snap_tasks.size()) {
+    /* length-sorted launch order for k_snappy (see TickPlan::d_snap) */
+    std::vector<SnapTask> sorted(r->snap_tasks.begin() + first,
+                                 r->snap_tasks.begin() + first + n);
+    std::stable_sort(sorted.begin(), sorted.end(),
+                     [](const SnapTask &a, const SnapTask &b) {
+                       return a.comp_len < b.comp_len;
+                     });
+    if (hipMalloc(&plan.d_snap, n * sizeof(SnapTask)) == hipSuccess &&
+        hipMemcpy(plan.d_snap, sorted.data(), n * sizeof(SnapTask),
+                  hipMemcpyHostToDevice) == hipSuccess) {
+      /* cached */
+    } else {
+      if (plan.d_snap) (void)hipFree(plan.d_snap);
+      plan.d_snap = nullptr; /* fall back to the unsorted full array */
+    }
+  }
   return r->plans.emplace(key, std::move(plan)).first->second;
 }
 
@@ -1921,7 +1947,9 @@ int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
   return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n,
                          plan.groups, plan.blob_bytes, false, nullptr, 0,
                          nullptr, nullptr, plan.d_groups, r->d_comp,
-                         r->snappy ? r->d_snaptasks + first : nullptr,
+                         !r->snappy ? nullptr
+                         : plan.d_snap ? plan.d_snap
+                                       : r->d_snaptasks + first,
                          std::move(counts));
 }
 
