@@ -1914,7 +1914,11 @@ static TickPlan &plan_for(GraReplay *r, uint64_t first, uint64_t n) {
   } else {
     plan.d_groups = nullptr; /* fall back to per-tick upload */
   }
-  if (r->snappy && first + n <= r->snap_tasks.size()) {
+  static const bool sort_snap = [] { /* A/B escape hatch */
+    const char *v = getenv("GRA_SNAPPY_SORT");
+    return !v || v[0] != '0';
+  }();
+  if (r->snappy && sort_snap && first + n <= r->snap_tasks.size()) {
     /* length-sorted launch order for k_snappy (see TickPlan::d_snap) */
     std::vector<SnapTask> sorted(r->snap_tasks.begin() + first,
                                  r->snap_tasks.begin() + first + n);
