@@ -1119,6 +1119,7 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     for (uint32_t g = 0; g < t.ngroups; g++) {
       DevRunDesc &rd = sl.h_rundescs[g];
       ShardState &ss = shards[rd.shard];
+      std::lock_guard<std::mutex> lk(ss.mu);
       if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
       stats.records += rd.n_entries;
       stats.payload_bytes += rd.payload_bytes;
