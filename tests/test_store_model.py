@@ -47,6 +47,8 @@ class ModelStore:
     def apply(self, ops):
         seq = self.latest + 1
         for kind, a, b in ops:
+            if kind == "log_data":  # WAL-only marker: no seq, no memtable
+                continue
             if kind == "delete_range":
                 self.tombs.append((seq, a, b))
             else:
