@@ -65,7 +65,7 @@ def build(ops, seq=0):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--cases", type=int, default=20000)
-    ap.add_argument("--seed", type=int, default=0xF022)
+    ap.add_argument("--seed", type=lambda s: int(s, 0), default=0xF022)
     args = ap.parse_args()
     rng = random.Random(args.seed)
     lib = oracle_ffi.load()
