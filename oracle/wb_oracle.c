@@ -405,10 +405,23 @@ int orc_apply(OrcStore *s, uint32_t shard, const uint8_t *rep, size_t len, int64
       }
       RangeTomb *rt = &t->tombs[t->ntombs++];
       rt->seq = seq;
-      rt->b_len = r->key_len;
-      rt->e_len = r->val_len;
-      rt->b_off = arena_add(t, rep + r->key_off, r->key_len);
-      rt->e_off = arena_add(t, rep + r->val_off, r->val_len);
+      if (r->cf_id != 0) {
+        /* CF range tombstones are namespaced like CF point keys: prefix the
+         * cf id to BOTH begin and end so covering comparisons against
+         * cf-prefixed query keys stay consistent (mirrors the engine's
+         * k_emit/host_build_run) */
+        rt->b_len = 4 + r->key_len;
+        rt->e_len = 4 + r->val_len;
+        rt->b_off = arena_add(t, (const uint8_t *)&r->cf_id, 4);
+        (void)arena_add(t, rep + r->key_off, r->key_len);
+        rt->e_off = arena_add(t, (const uint8_t *)&r->cf_id, 4);
+        (void)arena_add(t, rep + r->val_off, r->val_len);
+      } else {
+        rt->b_len = r->key_len;
+        rt->e_len = r->val_len;
+        rt->b_off = arena_add(t, rep + r->key_off, r->key_len);
+        rt->e_off = arena_add(t, rep + r->val_off, r->val_len);
+      }
     } else {
       /* CF-qualified keys are namespaced by prefixing the cf id (our store has
        * no column families; synthetic streams use cf 0). */

@@ -270,12 +270,18 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
   uint32_t rec = base.x;
   auto emit_one = [&](const wb::Rec &r, uint32_t idx) {
     uint32_t cf4 = r.cf_id ? 4u : 0u;
+    /* CF range tombstones prefix BOTH slices: the end key (value slice)
+     * lives in the same cf-namespaced key space as the begin key, so
+     * coverage comparisons against cf-prefixed query keys stay consistent
+     * (mirrored in host_build_run and the oracle) */
+    uint8_t btag = wb::base_tag(r.tag);
+    uint32_t cfv = (cf4 && btag == wb::kRangeDeletion) ? 4u : 0u;
     wb::RecHdr h;
     h.seq = d.base_seq + idx;
     h.kv_off = pay;
-    h.val_len = r.val_len;
+    h.val_len = r.val_len + cfv;
     h.key_len = (uint16_t)(r.key_len + cf4);
-    h.type = wb::base_tag(r.tag);
+    h.type = btag;
     h.flags = cf4 ? 1 : 0;
     h.kpref = 0; /* reserved; a key-prefix filter here measured NEGATIVE:
                     emit would re-touch blob key bytes (+35us/tick cold)
@@ -289,11 +295,18 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     tk.dst_rel = pay + cf4;
     tk.nbytes = r.key_len;
     tasks[2 * (rec + idx)] = tk;
+    if (cfv) { /* unaligned position: byte stores */
+      uint8_t *p = pay_region + pay + cf4 + r.key_len;
+      p[0] = (uint8_t)r.cf_id;
+      p[1] = (uint8_t)(r.cf_id >> 8);
+      p[2] = (uint8_t)(r.cf_id >> 16);
+      p[3] = (uint8_t)(r.cf_id >> 24);
+    }
     tk.src_off = d.off + r.val_off;
-    tk.dst_rel = pay + cf4 + r.key_len;
+    tk.dst_rel = pay + cf4 + r.key_len + cfv;
     tk.nbytes = r.val_len;
     tasks[2 * (rec + idx) + 1] = tk;
-    pay += (cf4 + r.key_len + r.val_len + 15u) & ~15u;
+    pay += (cf4 + r.key_len + cfv + r.val_len + 15u) & ~15u;
   };
   if (t.n_records <= kRecCache) { /* decode cached these — no blob re-walk */
     const wb::Rec *cache = reccache + (size_t)i * kRecCache;
@@ -1053,19 +1066,29 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
         }
         continue;
       }
+      if (ss.poisoned) {
+        /* shard already failed (earlier this tick or a previous tick):
+         * everything staged after the failure is stale — drop the group
+         * without touching durable_seq (rd.base_seq-1 may sit past a seq
+         * hole), keep next_seq at the durable boundary, prune the log */
+        ss.next_seq = ss.durable_seq + 1;
+        while (!ss.log.empty() && ss.log.back().base_seq > ss.durable_seq) {
+          ss.log_used -= ss.log.back().rep.size();
+          ss.log.pop_back();
+        }
+        continue;
+      }
       uint32_t keep_recs = 0;
       uint64_t keep_seq = rd.base_seq - 1;
-      bool bad = ss.poisoned; /* shard already failed earlier this tick */
-      if (!bad) {
-        for (uint32_t i = 0; i < gd.n_upds; i++) {
-          if (!sl.h_ok[gd.first + i]) {
-            bad = true;
-            break;
-          }
-          uint32_t c = have_counts ? t.counts[gd.first + i] : 0;
-          keep_recs += c;
-          keep_seq += c;
+      bool bad = false;
+      for (uint32_t i = 0; i < gd.n_upds; i++) {
+        if (!sl.h_ok[gd.first + i]) {
+          bad = true;
+          break;
         }
+        uint32_t c = have_counts ? t.counts[gd.first + i] : 0;
+        keep_recs += c;
+        keep_seq += c;
       }
       if (!bad) { /* whole group clean */
         auto run = std::make_shared<Run>();
@@ -1113,6 +1136,22 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     sl.busy = false;
     return GRA_OK;
   }
+  /* Stale-group guard for the err==0 paths: a tick staged BEFORE a corrupt
+   * batch's failure was detected carries base_seqs that assumed the corrupt
+   * batch applied. Such ticks always ingest while the shard is still
+   * poisoned (gra_handle_replicate_response's failure report drains pending
+   * ticks before clearing the flag), so drop the group, keep next_seq at
+   * the durable boundary and prune retained-log entries past it — the same
+   * handling as the error path's 'already failed earlier' case. Without
+   * this, durable_seq re-advances past the rolled-back boundary and the
+   * failed batch is silently skipped on re-pull (follower divergence). */
+  auto drop_stale_locked = [&](ShardState &ss) {
+    ss.next_seq = ss.durable_seq + 1;
+    while (!ss.log.empty() && ss.log.back().base_seq > ss.durable_seq) {
+      ss.log_used -= ss.log.back().rep.size();
+      ss.log.pop_back();
+    }
+  };
   if (opts.store_ring) {
     /* throughput store: runs are recycled by the ring; keep only seq/stats
      * bookkeeping (Get over recycled regions is undefined by contract) */
@@ -1120,6 +1159,10 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
       DevRunDesc &rd = sl.h_rundescs[g];
       ShardState &ss = shards[rd.shard];
       std::lock_guard<std::mutex> lk(ss.mu);
+      if (ss.poisoned) {
+        drop_stale_locked(ss);
+        continue;
+      }
       if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
       stats.records += rd.n_entries;
       stats.payload_bytes += rd.payload_bytes;
@@ -1132,6 +1175,10 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     DevRunDesc &rd = sl.h_rundescs[g];
     ShardState &ss = shards[rd.shard];
     std::lock_guard<std::mutex> lk(ss.mu);
+    if (ss.poisoned) {
+      drop_stale_locked(ss);
+      continue;
+    }
     auto run = std::make_shared<Run>();
     run->base_seq = rd.base_seq;
     run->last_seq = rd.last_seq;
@@ -1307,17 +1354,32 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
   ShardState &ss = e->shards[db->shard];
   uint64_t base;
   uint32_t count;
+  bool was_poisoned = false;
   {
     std::lock_guard<std::mutex> lk(ss.mu);
-    if (ss.poisoned) { /* reference cadence: fail once, caller re-pulls from
-                        * LatestSequenceNumber (replicated_db.cpp:378-382) */
-      ss.poisoned = false;
-      return 0;
+    if (ss.poisoned) {
+      was_poisoned = true;
+    } else {
+      if (len < wb::kHeaderBytes) return 0;
+      count = wb::fixed32_le(rep + 8);
+      base = ss.next_seq;
+      ss.next_seq += count; /* optimistic; rolled back on ANY failure below */
     }
-    if (len < wb::kHeaderBytes) return 0;
-    count = wb::fixed32_le(rep + 8);
-    base = ss.next_seq;
-    ss.next_seq += count; /* optimistic; rolled back on ANY failure below */
+  }
+  if (was_poisoned) {
+    /* Reference cadence: fail once, caller re-pulls from
+     * LatestSequenceNumber (replicated_db.cpp:378-382). Before clearing the
+     * flag, drain every pending tick: updates staged between the corrupt
+     * batch and its detection carry stale base_seqs, and the ingest paths
+     * drop their groups only while the shard is still poisoned. Error path
+     * only — never taken on healthy shards. */
+    {
+      std::lock_guard<std::mutex> lk(e->mu);
+      (void)e->flush_locked();
+    }
+    std::lock_guard<std::mutex> lk(ss.mu);
+    ss.poisoned = false;
+    return 0;
   }
   auto rollback = [&] {
     std::lock_guard<std::mutex> lk(ss.mu);

@@ -100,10 +100,15 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
     std::vector<wb::Rec> recs(tot.n_records);
     wb::WalkTotals t2 = wb::walk(rep, (uint32_t)len, recs.data(), tot.n_records);
     (void)t2;
+    /* CF range tombstones prefix BOTH slices (begin and end key live in the
+     * cf-namespaced key space) — mirrors k_emit and the oracle */
+    auto cfv_of = [](const wb::Rec &rc) {
+      return (rc.cf_id && wb::base_tag(rc.tag) == wb::kRangeDeletion) ? 4u : 0u;
+    };
     for (i = 0; i < tot.n_records; i++) {
       const wb::Rec &rc = recs[i];
       uint32_t cf4 = rc.cf_id ? 4u : 0u;
-      pay_need += (cf4 + rc.key_len + rc.val_len + 15u) & ~15u;
+      pay_need += (cf4 + rc.key_len + cfv_of(rc) + rc.val_len + 15u) & ~15u;
     }
     out->payload.resize(pay_need);
     out->payload_bytes = (uint32_t)pay_need;
@@ -112,10 +117,11 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
     for (i = 0; i < tot.n_records; i++) {
       const wb::Rec &rc = recs[i];
       uint32_t cf4 = rc.cf_id ? 4u : 0u;
+      uint32_t cfv = cfv_of(rc);
       wb::RecHdr h;
       h.seq = base_seq + i;
       h.kv_off = off;
-      h.val_len = rc.val_len;
+      h.val_len = rc.val_len + cfv;
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
@@ -124,8 +130,9 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);
       memcpy(p + cf4, rep + rc.key_off, rc.key_len);
-      memcpy(p + cf4 + rc.key_len, rep + rc.val_off, rc.val_len);
-      off += (cf4 + rc.key_len + rc.val_len + 15u) & ~15u;
+      if (cfv) memcpy(p + cf4 + rc.key_len, &rc.cf_id, 4);
+      memcpy(p + cf4 + rc.key_len + cfv, rep + rc.val_off, rc.val_len);
+      off += (cf4 + rc.key_len + cfv + rc.val_len + 15u) & ~15u;
     }
   }
   return true;

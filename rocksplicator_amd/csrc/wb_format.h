@@ -186,10 +186,14 @@ WB_HD WalkTotals walk_f(const uint8_t *rep, uint32_t len, F &&f) {
       r._pad = 0;
       r.cf_id = cf;
       f(r, nrec);
-      /* payload stored per record: [cf_id?4B] key value, 16-B aligned */
+      /* payload stored per record: [cf_id?4B] key [cf_id?4B-for-range-end]
+       * value, 16-B aligned. CF range tombstones prefix BOTH slices (begin
+       * AND end key live in the cf-namespaced key space) so covering
+       * comparisons against cf-prefixed query keys stay consistent. */
       uint32_t cf4 = cf ? 4u : 0u;
-      payload += slen[0] + slen[1] + cf4;
-      payload16 += (slen[0] + slen[1] + cf4 + 15u) & ~15u;
+      uint32_t cfx = (cf && bt == kRangeDeletion) ? 8u : cf4;
+      payload += slen[0] + slen[1] + cfx;
+      payload16 += (slen[0] + slen[1] + cfx + 15u) & ~15u;
       nrec++;
       consumed++;
     }

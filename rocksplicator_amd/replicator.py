@@ -56,6 +56,12 @@ class ReplicatedShard:
         self.upstream_db = upstream_db
         self._stop = threading.Event()
         self._thread = None
+        # write-degradation state is PER ReplicatedDB in the reference
+        # (replicated_db.cpp:236-273 keeps it on the instance): one shard's
+        # healthy acks must not reset another's degradation, and concurrent
+        # write() calls on this shard serialize the counter under a lock
+        self._miss_mu = threading.Lock()
+        self._consecutive_misses = 0
 
 
 class Replicator:
@@ -81,7 +87,6 @@ class Replicator:
         self._mu = threading.Lock()
         self._dbs = {}
         self._next_shard = 0
-        self._consecutive_misses = 0  # write-degradation counter
 
     # ≅ RocksDBReplicator::addDB (rocksdb_replicator.cpp:96-133)
     def add_db(self, name, role, upstream_db=None):
@@ -126,14 +131,16 @@ class Replicator:
             raise WriteToSlaveError(name)  # :107-109
         seq = rs.db.write_leader(rep_bytes)
         if mode in (1, 2):
-            timeout = (self.DEGRADED_TIMEOUT_MS
-                       if self._consecutive_misses >= self.DEGRADE_AFTER_MISSES
+            with rs._miss_mu:
+                degraded = rs._consecutive_misses >= self.DEGRADE_AFTER_MISSES
+            timeout = (self.DEGRADED_TIMEOUT_MS if degraded
                        else self.ACK_TIMEOUT_MS)
             ok = rs.db.wait_ack(seq, confirmed=(mode == 2), timeout_ms=timeout)
-            if ok:
-                self._consecutive_misses = 0
-            else:
-                self._consecutive_misses += 1  # degradation (:236-273)
+            with rs._miss_mu:
+                if ok:
+                    rs._consecutive_misses = 0
+                else:
+                    rs._consecutive_misses += 1  # degradation (:236-273)
         return seq
 
     def _start_pull(self, rs):

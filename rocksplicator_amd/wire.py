@@ -18,6 +18,7 @@ import socket
 import socketserver
 import struct
 import threading
+import time
 
 MAGIC = 0x47524150  # "GRAP"
 ROLE_FOLLOWER = 0
@@ -92,13 +93,21 @@ class UpdateServer:
                             self.request.sendall(encode_response([]))
                             continue
                         obs = _role == ROLE_OBSERVER
-                        ups = db.get_updates(seq_no, max_updates,
-                                             observer=obs)
-                        if not ups and max_wait_ms:
-                            with outer._cond:  # long-poll
-                                outer._cond.wait(max_wait_ms / 1e3)
-                            ups = db.get_updates(seq_no, max_updates,
-                                                 observer=obs)
+                        # long-poll: re-check UNDER the condition lock so a
+                        # notify_write between check and wait cannot be lost
+                        # (the reference's NonBlockingConditionVariable has
+                        # no such window)
+                        deadline = time.monotonic() + max_wait_ms / 1e3
+                        with outer._cond:
+                            while True:
+                                ups = db.get_updates(seq_no, max_updates,
+                                                     observer=obs)
+                                if ups:
+                                    break
+                                remaining = deadline - time.monotonic()
+                                if remaining <= 0:
+                                    break
+                                outer._cond.wait(remaining)
                         self.request.sendall(encode_response(ups))
                 except (ConnectionError, OSError):
                     pass

@@ -85,6 +85,17 @@ class PyBatch:
         self.count += 1
         return self
 
+    def cf_single_delete(self, cf, k):
+        self.body += bytes([TYPE_CF_SINGLE_DELETION]) + varint32(cf) + lps(k)
+        self.count += 1
+        return self
+
+    def cf_delete_range(self, cf, bk, ek):
+        self.body += (bytes([TYPE_CF_RANGE_DELETION]) + varint32(cf)
+                      + lps(bk) + lps(ek))
+        self.count += 1
+        return self
+
     def noop(self):
         self.body += bytes([TYPE_NOOP])  # consumes no seq
         return self
