@@ -456,7 +456,10 @@ def test_fuzz_all_record_types(olib):
             elif r < 0.70:
                 lo, hi = sorted([rng.choice(keys), rng.choice(keys)])
                 if lo != hi:
-                    b.delete_range(lo, hi)
+                    if rng.random() < 0.3:
+                        b.cf_delete_range(rng.randrange(1, 4), lo, hi)
+                    else:
+                        b.delete_range(lo, hi)
             elif r < 0.78:
                 b.cf_put(rng.randrange(1, 4), k, rng.randbytes(16))
             elif r < 0.83:
