@@ -30,8 +30,16 @@ rocksplicator_amd/libgra.so: $(OBJS)
 oracle/libwb_oracle.so: oracle/wb_oracle.c oracle/wb_oracle.h
 	$(MAKE) -C oracle
 
-# auxiliary binaries (C++ chain test, streaming bench, kernel microbenches)
-tools: build/test_cpp_chain build/bench_stream build/micro_copy build/micro_snappy
+# auxiliary binaries (C++ chain test, streaming bench, kernel microbenches,
+# DbWrapper-adapter compile proof)
+tools: build/test_cpp_chain build/bench_stream build/micro_copy build/micro_snappy build/seam_compile_check
+
+# compile-proof of INTEGRATION.md's GpuApplyDbWrapper against the exact
+# DbWrapper virtual seam (mock restatement of db_wrapper.h:6-15); plain g++
+# on purpose — the adapter is host C++ a maintainer builds without hipcc
+build/seam_compile_check: tools/mock_seam/compile_check.cpp tools/mock_seam/gpu_apply_db_wrapper.h tools/mock_seam/rocksdb_replicator/db_wrapper.h include/rocksplicator_gpu.h rocksplicator_amd/libgra.so | build
+	g++ -O2 -std=c++17 -Wall -Wextra $< -Itools/mock_seam -Iinclude -Lrocksplicator_amd -lgra -Wl,-rpath,'$$ORIGIN/../rocksplicator_amd' -o $@
+	./build/seam_compile_check
 
 build/test_cpp_chain: scripts/test_cpp_chain.cpp include/rocksplicator_replicator.hpp include/rocksplicator_gpu.h rocksplicator_amd/libgra.so | build
 	$(HIPCC) $(HIPFLAGS) $< -Iinclude -Lrocksplicator_amd -lgra -Wl,-rpath,'$$ORIGIN/../rocksplicator_amd' -o $@

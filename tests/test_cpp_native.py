@@ -31,3 +31,19 @@ def test_cpp_chain_runs():
     r = subprocess.run([BIN], capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "cpp chain OK" in r.stdout
+
+
+def test_seam_adapter_compiles_and_links():
+    """INTEGRATION.md's GpuApplyDbWrapper is a BUILT object, not prose:
+    compile it against the mock restatement of the reference's 4-method
+    DbWrapper seam (db_wrapper.h:6-15) and link against libgra.so, then run
+    the (GPU-free) link proof."""
+    out = os.path.join(REPO, "build", "seam_compile_check")
+    subprocess.run(
+        ["g++", "-O2", "-std=c++17", "-Wall", "-Wextra",
+         "tools/mock_seam/compile_check.cpp", "-Itools/mock_seam", "-Iinclude",
+         "-Lrocksplicator_amd", "-lgra",
+         "-Wl,-rpath,$ORIGIN/../rocksplicator_amd", "-o", out],
+        cwd=REPO, check=True)
+    r = subprocess.run([out], capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0 and "adapter compile check OK" in r.stdout
