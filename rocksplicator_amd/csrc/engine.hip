@@ -636,6 +636,10 @@ struct GraEngine {
   GraEngineOpts opts;
   hipStream_t stream = nullptr;      /* pipeline kernels */
   hipStream_t copyout = nullptr;     /* run-descriptor D2H, overlapped */
+  hipStream_t h2d = nullptr;         /* PCIe staging H2D, overlapped: tick
+                                        N+1's copy runs under tick N's
+                                        kernels (double-buffered device
+                                        staging below) */
   /* device store */
   uint8_t *d_store = nullptr;
   uint64_t *d_cursor = nullptr;
@@ -669,6 +673,15 @@ struct GraEngine {
   };
   StageBuf stage[2];
   std::atomic<StageBuf *> cur_stage{nullptr};
+  /* double-buffered device staging: staged ticks alternate buffers so the
+   * H2D into buffer B (h2d stream) overlaps the kernels still reading
+   * buffer A (main stream). stage_used_ev[b] is recorded on the main
+   * stream after the last kernel reading buffer b; the next H2D into b
+   * waits on it. d_stage_blobs/d_stage_descs alias buffer 0 (capacity
+   * checks + call-site compatibility). */
+  uint8_t *d_stage_blobs_bufs[2] = {nullptr, nullptr};
+  UpdDesc *d_stage_descs_bufs[2] = {nullptr, nullptr};
+  hipEvent_t stage_used_ev[2] = {nullptr, nullptr};
   uint8_t *d_stage_blobs = nullptr;
   UpdDesc *d_stage_descs = nullptr;
   /* slots + pending ticks */
@@ -782,6 +795,7 @@ int GraEngine::init(const GraEngineOpts &o) {
   }
   HIP_TRY(hipStreamCreate(&stream));
   HIP_TRY(hipStreamCreate(&copyout));
+  HIP_TRY(hipStreamCreate(&h2d));
   HIP_TRY(hipMalloc(&d_store, opts.store_bytes + 16));
   HIP_TRY(hipMalloc(&d_cursor, 8));
   HIP_TRY(hipMemset(d_cursor, 0, 8));
@@ -794,8 +808,14 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMalloc(&d_groups, (size_t)group_cap * sizeof(GroupDesc)));
   HIP_TRY(hipMalloc(&d_err_ring, kErrRing * 4));
   HIP_TRY(hipMemset(d_err_ring, 0, kErrRing * 4));
-  HIP_TRY(hipMalloc(&d_stage_blobs, opts.staging_bytes + 16));
-  HIP_TRY(hipMalloc(&d_stage_descs, (size_t)max_upd * sizeof(UpdDesc)));
+  for (int i = 0; i < 2; i++) {
+    HIP_TRY(hipMalloc(&d_stage_blobs_bufs[i], opts.staging_bytes + 16));
+    HIP_TRY(hipMalloc(&d_stage_descs_bufs[i], (size_t)max_upd * sizeof(UpdDesc)));
+    HIP_TRY(hipEventCreate(&stage_used_ev[i]));
+    HIP_TRY(hipEventRecord(stage_used_ev[i], stream));
+  }
+  d_stage_blobs = d_stage_blobs_bufs[0];
+  d_stage_descs = d_stage_descs_bufs[0];
   /* epochs must be unique ACROSS engine instances: a thread-local staging
    * chunk could otherwise match a freshly created engine reusing the same
    * heap address (ABA) */
@@ -826,6 +846,7 @@ GraEngine::~GraEngine() {
   (void)hipStreamSynchronize(stream);
   (void)hipStreamSynchronize(copyout); /* pending rundesc/ok D2H target the
                                           pinned slot buffers freed below */
+  (void)hipStreamSynchronize(h2d);     /* pending H2D reads pinned staging */
   for (auto &t : pending)
     for (int i = 0; i < kEventsPerTick; i++)
       if (t.ev[i]) (void)hipEventDestroy(t.ev[i]);
@@ -848,14 +869,19 @@ GraEngine::~GraEngine() {
   for (void *p : {(void *)mg.d_runs, (void *)mg.d_keys, (void *)mg.d_keybuf,
                   (void *)mg.d_valbuf, (void *)mg.d_out})
     if (p) (void)hipFree(p);
+  for (int i = 0; i < 2; i++) {
+    if (d_stage_blobs_bufs[i]) (void)hipFree(d_stage_blobs_bufs[i]);
+    if (d_stage_descs_bufs[i]) (void)hipFree(d_stage_descs_bufs[i]);
+    if (stage_used_ev[i]) (void)hipEventDestroy(stage_used_ev[i]);
+  }
   for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_totals,
                   (void *)d_partial, (void *)d_bsums, (void *)d_reccache,
                   (void *)d_tasks, (void *)d_place, (void *)d_groups,
-                  (void *)d_err_ring, (void *)d_stage_blobs,
-                  (void *)d_stage_descs})
+                  (void *)d_err_ring})
     if (p) (void)hipFree(p);
   if (stream) (void)hipStreamDestroy(stream);
   if (copyout) (void)hipStreamDestroy(copyout);
+  if (h2d) (void)hipStreamDestroy(h2d);
 }
 
 int GraEngine::free_slot() {
@@ -924,17 +950,33 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                            hipMemcpyHostToDevice, stream));
   }
   HIP_TRY(rec(0)); /* tick start */
-  if (h2d_src != nullptr) { /* PCIe-inclusive path: stage blobs (+descs) */
+  int stage_buf = -1;
+  if (h2d_src != nullptr) { /* PCIe-inclusive path: stage blobs (+descs) on
+                             * the dedicated h2d stream, double-buffered —
+                             * this tick's copy overlaps the previous tick's
+                             * kernels (which read the other buffer) */
+    stage_buf = (int)(tick & 1u);
+    if (d_h2d_dst == d_stage_blobs_bufs[0]) { /* engine-managed staging */
+      d_h2d_dst = d_stage_blobs_bufs[stage_buf];
+      d_blobs = d_stage_blobs_bufs[stage_buf];
+      if (h_descs_src) d_descw = d_stage_descs_bufs[stage_buf];
+    }
+    /* previous reader of this buffer must be done before overwrite */
+    HIP_TRY(hipStreamWaitEvent(h2d, stage_used_ev[stage_buf], 0));
+    t.evmask |= 1u << 9; /* h2d leg timed on its own stream: ev9 -> ev1 */
+    HIP_TRY(hipEventRecord(t.ev[9], h2d));
     HIP_TRY(hipMemcpyAsync(d_h2d_dst, h2d_src, h2d_bytes,
-                           hipMemcpyHostToDevice, stream));
+                           hipMemcpyHostToDevice, h2d));
     if (h_descs_src) {
       memcpy(sl.h_descs, h_descs_src, (size_t)n * sizeof(UpdDesc));
       HIP_TRY(hipMemcpyAsync((void *)d_descw, sl.h_descs,
                              (size_t)n * sizeof(UpdDesc),
-                             hipMemcpyHostToDevice, stream));
+                             hipMemcpyHostToDevice, h2d));
     }
+    t.evmask |= 1u << 1;
+    HIP_TRY(hipEventRecord(t.ev[1], h2d));
+    HIP_TRY(hipStreamWaitEvent(stream, t.ev[1], 0)); /* kernels gate on data */
   }
-  if (h2d_src != nullptr) HIP_TRY(rec(1)); /* after h2d */
   uint32_t nb = (n + 255) / 256;
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena */
     hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, stream, d_comp,
@@ -972,6 +1014,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                      d_partial, d_bsums, n, nb, d_place, sl.d_rundescs);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(6)); /* main-stream tick end */
+  if (stage_buf >= 0) /* last reader of this staging buffer has retired */
+    HIP_TRY(hipEventRecord(stage_used_ev[stage_buf], stream));
   /* publish run descriptors on the copyout stream, overlapped with the next
    * tick's kernels (per-slot device buffer: no hazard with slot reuse —
    * ingest waits on ev[7], which gates sl.busy) */
@@ -1013,7 +1057,8 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
   };
   int prev = 0;
   if (has(1)) {
-    stats.h2d_ms += dt(0, 1);
+    stats.h2d_ms += dt(has(9) ? 9 : 0, 1); /* ev9->ev1 = copy on the h2d
+                                              stream (overlapped path) */
     prev = 1;
   }
   if (has(8)) {
@@ -1284,8 +1329,9 @@ int GraEngine::stream_tick_locked() {
                         true, old->pin, fill, d_stage_blobs, ud.data(), nullptr,
                         nullptr, nullptr, std::move(counts));
   if (rc != GRA_OK) return rc;
-  /* mark the old pinned buffer reusable once its H2D completed */
-  HIP_TRY(hipEventRecord(old->free_ev, stream));
+  /* mark the old pinned buffer reusable once its H2D completed (the copy
+   * runs on the dedicated h2d stream now) */
+  HIP_TRY(hipEventRecord(old->free_ev, h2d));
   return GRA_OK;
 }
 

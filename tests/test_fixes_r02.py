@@ -119,7 +119,5 @@ def test_cf_range_delete_checksum_stable(olib):
     for _ in range(2):
         st = oracle_ffi.Store(olib, 1)
         assert st.apply(0, rep)
-        s = C.c_uint64()
-        olib.orc_shard_checksum(st.h, 0, C.byref(s))
-        sums.append(s.value)
-    assert sums[0] == sums[1]
+        sums.append(olib.orc_shard_checksum(st.h, 0))
+    assert sums[0] == sums[1] != 0

@@ -866,9 +866,7 @@ def test_stale_tick_after_corruption_dropped(olib):
 
 
 def olib_checksum(olib, ost, shard=0):
-    v = C.c_uint64()
-    olib.orc_shard_checksum(ost.h, shard, C.byref(v))
-    return v.value
+    return olib.orc_shard_checksum(ost.h, shard)
 
 
 def test_stale_streaming_ticks_after_corruption(olib):
