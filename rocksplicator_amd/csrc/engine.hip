@@ -682,6 +682,12 @@ struct GraEngine {
   uint8_t *d_stage_blobs_bufs[2] = {nullptr, nullptr};
   UpdDesc *d_stage_descs_bufs[2] = {nullptr, nullptr};
   hipEvent_t stage_used_ev[2] = {nullptr, nullptr};
+  /* host-side submission gate: at most 2 staged copies queued on the h2d
+   * stream. Measured on MI355X (profiles/r02): with >4 queued async
+   * copies the submitting thread blocks INSIDE hipMemcpyAsync and the
+   * in-flight SDMA transfers degrade 56 -> 37 GB/s; blocking the host in
+   * hipEventSynchronize instead keeps them at full rate. */
+  hipEvent_t h2d_gate_ev[2] = {nullptr, nullptr};
   uint8_t *d_stage_blobs = nullptr;
   UpdDesc *d_stage_descs = nullptr;
   /* slots + pending ticks */
@@ -730,6 +736,11 @@ struct TickPlan {
   std::vector<GroupDesc> groups;
   uint64_t blob_bytes = 0;
   GroupDesc *d_groups = nullptr; /* device-cached copy (freed with replay) */
+  UpdDesc *d_ud = nullptr;       /* tick_h2d: window descs REBASED to the
+                                    staged window, cached on device — the
+                                    per-step 20 MB desc rebase + H2D that
+                                    throttled the staged leg to ~39 GB/s
+                                    happens once per window instead */
   SnapTask *d_snap = nullptr;    /* window's snap tasks, LENGTH-SORTED so a
                                     wave's lanes get similar-size streams
                                     (+15% k_snappy, scripts/micro_snappy.hip
@@ -813,6 +824,8 @@ int GraEngine::init(const GraEngineOpts &o) {
     HIP_TRY(hipMalloc(&d_stage_descs_bufs[i], (size_t)max_upd * sizeof(UpdDesc)));
     HIP_TRY(hipEventCreate(&stage_used_ev[i]));
     HIP_TRY(hipEventRecord(stage_used_ev[i], stream));
+    HIP_TRY(hipEventCreateWithFlags(&h2d_gate_ev[i], hipEventDisableTiming));
+    HIP_TRY(hipEventRecord(h2d_gate_ev[i], h2d));
   }
   d_stage_blobs = d_stage_blobs_bufs[0];
   d_stage_descs = d_stage_descs_bufs[0];
@@ -873,6 +886,7 @@ GraEngine::~GraEngine() {
     if (d_stage_blobs_bufs[i]) (void)hipFree(d_stage_blobs_bufs[i]);
     if (d_stage_descs_bufs[i]) (void)hipFree(d_stage_descs_bufs[i]);
     if (stage_used_ev[i]) (void)hipEventDestroy(stage_used_ev[i]);
+    if (h2d_gate_ev[i]) (void)hipEventDestroy(h2d_gate_ev[i]);
   }
   for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_totals,
                   (void *)d_partial, (void *)d_bsums, (void *)d_reccache,
@@ -921,11 +935,13 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   Slot &sl = slots[si];
   sl.busy = true;
   uint32_t ngroups = (uint32_t)groups.size();
+  /* the host copy is ALWAYS filled: ingest's corruption recovery walks
+   * sl.h_groups to find each group's update range (r01 bug: with a
+   * device-cached plan the host copy stayed stale and the error path read
+   * garbage groups — replay-tick corruption went undetected) */
+  memcpy(sl.h_groups, groups.data(), ngroups * sizeof(GroupDesc));
   const GroupDesc *groups_for_kernel = d_groups_dev;
-  if (!groups_for_kernel) {
-    memcpy(sl.h_groups, groups.data(), ngroups * sizeof(GroupDesc));
-    groups_for_kernel = d_groups;
-  }
+  if (!groups_for_kernel) groups_for_kernel = d_groups;
   uint32_t tick = tick_id++;
 
   TickRec t;
@@ -961,6 +977,10 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
       d_blobs = d_stage_blobs_bufs[stage_buf];
       if (h_descs_src) d_descw = d_stage_descs_bufs[stage_buf];
     }
+    /* host gate: the copy that last used this buffer (2 staged ticks ago)
+     * must have completed — bounds the SDMA queue to <=2 pending copies
+     * (see h2d_gate_ev above for why) */
+    HIP_TRY(hipEventSynchronize(h2d_gate_ev[stage_buf]));
     /* previous reader of this buffer must be done before overwrite */
     HIP_TRY(hipStreamWaitEvent(h2d, stage_used_ev[stage_buf], 0));
     t.evmask |= 1u << 9; /* h2d leg timed on its own stream: ev9 -> ev1 */
@@ -975,6 +995,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
     }
     t.evmask |= 1u << 1;
     HIP_TRY(hipEventRecord(t.ev[1], h2d));
+    HIP_TRY(hipEventRecord(h2d_gate_ev[stage_buf], h2d));
     HIP_TRY(hipStreamWaitEvent(stream, t.ev[1], 0)); /* kernels gate on data */
   }
   uint32_t nb = (n + 255) / 256;
@@ -1989,6 +2010,7 @@ void gra_replay_destroy(GraReplay *r) {
   for (auto &kv : r->plans) {
     if (kv.second.d_groups) (void)hipFree(kv.second.d_groups);
     if (kv.second.d_snap) (void)hipFree(kv.second.d_snap);
+    if (kv.second.d_ud) (void)hipFree(kv.second.d_ud);
   }
   delete r;
 }
@@ -2085,16 +2107,27 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
     return GRA_ERR;
   }
   TickPlan &plan = plan_for(r, first, n);
-  /* rebased descs (blob offsets relative to staged window) */
-  std::vector<UpdDesc> ud(n);
-  for (uint64_t i = 0; i < n; i++) {
-    ud[i] = r->descs[first + i];
-    ud[i].off -= lo;
-  }
   std::lock_guard<std::mutex> lk(e->mu);
-  return e->enqueue_tick(e->d_stage_blobs, e->d_stage_descs, (uint32_t)n,
+  if (!plan.d_ud) {
+    /* rebased descs (blob offsets relative to the staged window), uploaded
+     * once per window and reused every step */
+    std::vector<UpdDesc> ud(n);
+    for (uint64_t i = 0; i < n; i++) {
+      ud[i] = r->descs[first + i];
+      ud[i].off -= lo;
+    }
+    if (hipMalloc(&plan.d_ud, n * sizeof(UpdDesc)) != hipSuccess ||
+        hipMemcpy(plan.d_ud, ud.data(), n * sizeof(UpdDesc),
+                  hipMemcpyHostToDevice) != hipSuccess) {
+      if (plan.d_ud) (void)hipFree(plan.d_ud);
+      plan.d_ud = nullptr;
+      g_err = "tick_h2d: desc cache allocation failed";
+      return GRA_ERR;
+    }
+  }
+  return e->enqueue_tick(e->d_stage_blobs, plan.d_ud, (uint32_t)n,
                          plan.groups, plan.blob_bytes, true, r->h_arena + lo,
-                         hi - lo, e->d_stage_blobs, ud.data());
+                         hi - lo, e->d_stage_blobs, nullptr);
 }
 
 int gra_replay_sync(GraReplay *r) { return gra_flush(r->e); }
