@@ -275,6 +275,218 @@ __global__ void k_v3g(const uint8_t *__restrict__ comp,
   decompress_pat(comp + t.comp_off, t.comp_len, out + t.out_off, t.ulen);
 }
 
+/* v5: two-phase per-lane — phase 1 parses the element headers into compact
+ * u64 tokens (chunks of TCHUNK in LDS), phase 2 executes them with a
+ * branch-light copy ladder. Separates the dependent/divergent parse from
+ * the data movement. Valid when ulen <= 65535 (u16 token fields) and
+ * clen <= MAXC; otherwise falls back to the stock path. */
+template <int MAXC, int TCHUNK>
+__global__ void __launch_bounds__(256) k_v5(const uint8_t *__restrict__ comp,
+                                            const Task *__restrict__ tasks,
+                                            uint32_t n,
+                                            uint8_t *__restrict__ out) {
+  constexpr int STRIDE = MAXC + 4;
+  __shared__ uint8_t lds[256 * STRIDE];
+  __shared__ uint64_t tok[256 * TCHUNK];
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  Task t = tasks[i];
+  uint8_t *dst = out + t.out_off;
+  if (t.comp_len > MAXC || t.ulen > 65535) {
+    snp::decompress(comp + t.comp_off, t.comp_len, dst, t.ulen);
+    return;
+  }
+  uint8_t *mine = lds + threadIdx.x * STRIDE;
+  {
+    const uint8_t *gsrc = comp + t.comp_off;
+    for (uint32_t b = 0; b < t.comp_len; b += 16)
+      *(uint4 *)(mine + b) = *(const uint4 *)(gsrc + b);
+  }
+  uint64_t *mytok = tok + threadIdx.x * TCHUNK;
+  uint32_t slen = t.comp_len;
+  uint32_t ulen = 0;
+  uint32_t ip = wb::varint32(mine, slen, &ulen);
+  if (ip == 0 || ulen > t.ulen) return;
+  uint32_t op = 0;
+  bool bad = false;
+  while (ip < slen && !bad) {
+    uint32_t nt = 0;
+    while (ip < slen && nt < TCHUNK) { /* ---- phase 1: parse ---- */
+      uint8_t tag = mine[ip++];
+      if ((tag & 3) == 0) {
+        uint64_t len64 = (uint32_t)(tag >> 2) + 1;
+        if (len64 > 60) {
+          uint32_t nb = (uint32_t)len64 - 60;
+          if ((uint64_t)ip + nb > slen) { bad = true; break; }
+          len64 = 0;
+          for (uint32_t b = 0; b < nb; b++)
+            len64 |= (uint64_t)mine[ip + b] << (8 * b);
+          len64 += 1;
+          ip += nb;
+        }
+        if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen) {
+          bad = true;
+          break;
+        }
+        uint32_t len = (uint32_t)len64;
+        mytok[nt++] = (uint64_t)op | ((uint64_t)len << 16) | ((uint64_t)ip << 32);
+        ip += len;
+        op += len;
+      } else {
+        uint32_t len, off;
+        if ((tag & 3) == 1) {
+          len = ((tag >> 2) & 7) + 4;
+          if (ip + 1 > slen) { bad = true; break; }
+          off = ((uint32_t)(tag >> 5) << 8) | mine[ip];
+          ip += 1;
+        } else if ((tag & 3) == 2) {
+          len = (tag >> 2) + 1;
+          if (ip + 2 > slen) { bad = true; break; }
+          off = (uint32_t)mine[ip] | ((uint32_t)mine[ip + 1] << 8);
+          ip += 2;
+        } else {
+          len = (tag >> 2) + 1;
+          if (ip + 4 > slen) { bad = true; break; }
+          off = (uint32_t)mine[ip] | ((uint32_t)mine[ip + 1] << 8) |
+                ((uint32_t)mine[ip + 2] << 16) | ((uint32_t)mine[ip + 3] << 24);
+          ip += 4;
+        }
+        if (off == 0 || off > op || (uint64_t)op + len > ulen) {
+          bad = true;
+          break;
+        }
+        mytok[nt++] = (uint64_t)op | ((uint64_t)len << 16) |
+                      ((uint64_t)off << 32) | (1ull << 48);
+        op += len;
+      }
+    }
+    for (uint32_t k = 0; k < nt; k++) { /* ---- phase 2: execute ---- */
+      uint64_t tk = mytok[k];
+      uint32_t d0 = tk & 0xFFFF, len = (tk >> 16) & 0xFFFF,
+               s = (tk >> 32) & 0xFFFF;
+      if (!(tk >> 48)) { /* literal from the LDS stage */
+        for (uint32_t b = 0; b < len; b += 16)
+          *(uint4 *)(dst + d0 + b) = *(const uint4 *)(mine + s + b);
+      } else if (s >= len) {
+        for (uint32_t b = 0; b < len; b += 16)
+          *(uint4 *)(dst + d0 + b) = *(const uint4 *)(dst + d0 - s + b);
+      } else if (s == 16) { /* dominant periodic case: materialize */
+        uint4 P = *(const uint4 *)(dst + d0 - 16);
+        for (uint32_t b = 0; b < len; b += 16) *(uint4 *)(dst + d0 + b) = P;
+      } else if (s >= 8) {
+        for (uint32_t b = 0; b < len; b += 8)
+          *(uint64_t *)(dst + d0 + b) = *(const uint64_t *)(dst + d0 - s + b);
+      } else {
+        for (uint32_t b = 0; b < len; b++) dst[d0 + b] = dst[d0 + b - s];
+      }
+    }
+  }
+}
+
+/* v6: G-lane cooperative with per-update LDS staging. 256/G updates per
+ * block cuts static LDS to (256/G)*(MAXC+4) — 32 KB at G=4 — so several
+ * blocks fit a CU (v2's 128 KB pins occupancy at 1). All G lanes parse
+ * the same stream redundantly (converged control flow within the group);
+ * element payloads move with G parallel 16-B chunks. Overlapping copies
+ * read only pre-element bytes (period pattern), so lanes never race. */
+template <int G>
+__device__ uint32_t decomp_coopG(const uint8_t *__restrict__ src,
+                                 uint32_t slen, uint8_t *__restrict__ dst,
+                                 uint32_t dcap, uint32_t lane) {
+  uint32_t ulen = 0;
+  uint32_t ip = wb::varint32(src, slen, &ulen);
+  if (ip == 0 || ulen > dcap) return UINT32_MAX;
+  uint32_t op = 0;
+  while (ip < slen) {
+    uint8_t tag = src[ip++];
+    if ((tag & 3) == 0) { /* literal */
+      uint64_t len64 = (uint32_t)(tag >> 2) + 1;
+      if (len64 > 60) {
+        uint32_t nb = (uint32_t)len64 - 60;
+        if ((uint64_t)ip + nb > slen) return UINT32_MAX;
+        len64 = 0;
+        for (uint32_t b = 0; b < nb; b++)
+          len64 |= (uint64_t)src[ip + b] << (8 * b);
+        len64 += 1;
+        ip += nb;
+      }
+      if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen)
+        return UINT32_MAX;
+      uint32_t len = (uint32_t)len64;
+      for (uint32_t b = lane * 16; b < len; b += G * 16)
+        *(uint4 *)(dst + op + b) = *(const uint4 *)(src + ip + b);
+      ip += len;
+      op += len;
+    } else {
+      uint32_t len, off;
+      if ((tag & 3) == 1) {
+        len = ((tag >> 2) & 7) + 4;
+        if (ip + 1 > slen) return UINT32_MAX;
+        off = ((uint32_t)(tag >> 5) << 8) | src[ip];
+        ip += 1;
+      } else if ((tag & 3) == 2) {
+        len = (tag >> 2) + 1;
+        if (ip + 2 > slen) return UINT32_MAX;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+      } else {
+        len = (tag >> 2) + 1;
+        if (ip + 4 > slen) return UINT32_MAX;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+              ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > op || (uint64_t)op + len > ulen) return UINT32_MAX;
+      if (off >= len) { /* no overlap: parallel chunks */
+        for (uint32_t b = lane * 16; b < len; b += G * 16)
+          *(uint4 *)(dst + op + b) = *(const uint4 *)(dst + op - off + b);
+      } else if (off == 16) {
+        uint4 P = *(const uint4 *)(dst + op - 16); /* pre-element: safe */
+        for (uint32_t b = lane * 16; b < len; b += G * 16)
+          *(uint4 *)(dst + op + b) = P;
+      } else if (off == 8) {
+        uint64_t q = *(const uint64_t *)(dst + op - 8);
+        for (uint32_t b = lane * 8; b < len; b += G * 8)
+          *(uint64_t *)(dst + op + b) = q;
+      } else { /* general period: read pre-element bytes only */
+        for (uint32_t b = lane; b < len; b += G)
+          dst[op + b] = dst[op - off + (b % off)];
+      }
+      op += len;
+    }
+  }
+  return op == ulen ? op : UINT32_MAX;
+}
+
+template <int G, int MAXC>
+__global__ void __launch_bounds__(256) k_v6(const uint8_t *__restrict__ comp,
+                                            const Task *__restrict__ tasks,
+                                            uint32_t n,
+                                            uint8_t *__restrict__ out) {
+  constexpr int NUPD = 256 / G;
+  constexpr int STRIDE = MAXC + 4;
+  __shared__ uint8_t lds[NUPD * STRIDE];
+  uint32_t lane = threadIdx.x % G;
+  uint32_t u = threadIdx.x / G;
+  uint32_t i = blockIdx.x * NUPD + u;
+  bool active = i < n;
+  Task t{};
+  if (active) t = tasks[i];
+  bool staged = active && t.comp_len <= MAXC;
+  uint8_t *mine = lds + u * STRIDE;
+  if (staged) {
+    const uint8_t *src = comp + t.comp_off;
+    for (uint32_t b = lane * 16; b < t.comp_len; b += G * 16)
+      *(uint4 *)(mine + b) = *(const uint4 *)(src + b);
+  }
+  __syncthreads(); /* cross-lane LDS visibility within each group */
+  if (!active) return;
+  if (staged)
+    decomp_coopG<G>(mine, t.comp_len, out + t.out_off, t.ulen, lane);
+  else if (lane == 0)
+    snp::decompress(comp + t.comp_off, t.comp_len, out + t.out_off, t.ulen);
+}
+
 int main(int argc, char **argv) {
   uint32_t n = argc > 1 ? atoi(argv[1]) : 400000;
   uint32_t vlen = argc > 2 ? atoi(argv[2]) : 1024;
@@ -415,6 +627,41 @@ int main(int argc, char **argv) {
     run("v2-lensorted", [&] {
       hipLaunchKernelGGL(k_v2<508>, dim3((n + 255) / 256), dim3(256), 0, 0,
                          d_comp, d_sorted, n, d_out);
+    });
+    /* round-2 variants, sorted launch order (the adopted default) */
+    run("v5-2p-sorted", [&] {
+      hipLaunchKernelGGL((k_v5<508, 8>), dim3((n + 255) / 256), dim3(256), 0,
+                         0, d_comp, d_sorted, n, d_out);
+    });
+    run("v6-coop2-sort", [&] {
+      hipLaunchKernelGGL((k_v6<2, 508>), dim3(((size_t)n * 2 + 255) / 256),
+                         dim3(256), 0, 0, d_comp, d_sorted, n, d_out);
+    });
+    run("v6-coop4-sort", [&] {
+      hipLaunchKernelGGL((k_v6<4, 508>), dim3(((size_t)n * 4 + 255) / 256),
+                         dim3(256), 0, 0, d_comp, d_sorted, n, d_out);
+    });
+    run("v6-coop8-sort", [&] {
+      hipLaunchKernelGGL((k_v6<8, 508>), dim3(((size_t)n * 8 + 255) / 256),
+                         dim3(256), 0, 0, d_comp, d_sorted, n, d_out);
+    });
+    run("v5-2p-unsort", [&] {
+      hipLaunchKernelGGL((k_v5<508, 8>), dim3((n + 255) / 256), dim3(256), 0,
+                         0, d_comp, d_tasks, n, d_out);
+    });
+    run("v6-coop4-uns", [&] {
+      hipLaunchKernelGGL((k_v6<4, 508>), dim3(((size_t)n * 4 + 255) / 256),
+                         dim3(256), 0, 0, d_comp, d_tasks, n, d_out);
+    });
+    /* small-stage occupancy probe: sorted order means most blocks' streams
+     * fit 252 B and the tail falls back to the global parse */
+    run("v2-252-sort", [&] {
+      hipLaunchKernelGGL(k_v2<252>, dim3((n + 255) / 256), dim3(256), 0, 0,
+                         d_comp, d_sorted, n, d_out);
+    });
+    run("v6c4-252-sort", [&] {
+      hipLaunchKernelGGL((k_v6<4, 252>), dim3(((size_t)n * 4 + 255) / 256),
+                         dim3(256), 0, 0, d_comp, d_sorted, n, d_out);
     });
     CHECK(hipFree(d_sorted));
   }
