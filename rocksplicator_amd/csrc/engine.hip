@@ -644,12 +644,16 @@ struct GraEngine {
                                         tick N's decompress runs under tick
                                         N-1's decode..copy. Scratch slots
                                         are per-update (no aliasing across
-                                        windows); window reuse is fenced by
-                                        prep_gate_ev (recorded after each
-                                        tick's k_copy — the last blob
-                                        reader), so prep runs at most one
-                                        tick ahead of main. */
-  hipEvent_t prep_gate_ev = nullptr;
+                                        windows), so the only hazard is a
+                                        WINDOW being re-decompressed while
+                                        an older tick of the same window
+                                        still reads it — fenced by the
+                                        per-window consumed event the
+                                        caller passes (TickPlan::
+                                        consumed_ev), not by a global
+                                        gate (a global gate serializes
+                                        prep behind main and kills the
+                                        overlap). */
   /* device store */
   uint8_t *d_store = nullptr;
   uint64_t *d_cursor = nullptr;
@@ -727,7 +731,8 @@ struct GraEngine {
                    const GroupDesc *d_groups_dev = nullptr,
                    const uint8_t *d_comp = nullptr,
                    const SnapTask *d_snaptasks = nullptr,
-                   std::vector<uint32_t> &&counts = {});
+                   std::vector<uint32_t> &&counts = {},
+                   hipEvent_t window_ev = nullptr);
   int ingest(bool wait_all);
   int ingest_one(TickRec &t, bool wait);
   int flush_locked();
@@ -751,6 +756,11 @@ struct TickPlan {
                                     per-step 20 MB desc rebase + H2D that
                                     throttled the staged leg to ~39 GB/s
                                     happens once per window instead */
+  hipEvent_t consumed_ev = nullptr; /* snappy overlap: recorded after each
+                                       tick of this window finishes reading
+                                       its scratch slots (k_copy); the next
+                                       k_snappy of the SAME window waits on
+                                       it */
   SnapTask *d_snap = nullptr;    /* window's snap tasks, LENGTH-SORTED so a
                                     wave's lanes get similar-size streams
                                     (+15% k_snappy, scripts/micro_snappy.hip
@@ -818,7 +828,6 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipStreamCreate(&copyout));
   HIP_TRY(hipStreamCreate(&h2d));
   HIP_TRY(hipStreamCreate(&prep));
-  HIP_TRY(hipEventCreateWithFlags(&prep_gate_ev, hipEventDisableTiming));
   HIP_TRY(hipMalloc(&d_store, opts.store_bytes + 16));
   HIP_TRY(hipMalloc(&d_cursor, 8));
   HIP_TRY(hipMemset(d_cursor, 0, 8));
@@ -917,7 +926,6 @@ GraEngine::~GraEngine() {
   if (copyout) (void)hipStreamDestroy(copyout);
   if (h2d) (void)hipStreamDestroy(h2d);
   if (prep) (void)hipStreamDestroy(prep);
-  if (prep_gate_ev) (void)hipEventDestroy(prep_gate_ev);
 }
 
 int GraEngine::free_slot() {
@@ -942,7 +950,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                             const GroupDesc *d_groups_dev,
                             const uint8_t *d_comp,
                             const SnapTask *d_snaptasks,
-                            std::vector<uint32_t> &&counts) {
+                            std::vector<uint32_t> &&counts,
+                            hipEvent_t window_ev) {
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
@@ -1023,10 +1032,11 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   uint32_t nb = (n + 255) / 256;
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena
                       * on the prep stream, overlapped with the previous
-                      * tick's decode..copy. prep_gate_ev (recorded after
-                      * the previous k_copy, the last blob reader) fences
-                      * window reuse. */
-    HIP_TRY(hipStreamWaitEvent(prep, prep_gate_ev, 0));
+                      * tick's decode..copy. window_ev (recorded after the
+                      * LAST tick of this same window finished reading the
+                      * scratch) fences window reuse without serializing
+                      * prep behind main. */
+    if (window_ev) HIP_TRY(hipStreamWaitEvent(prep, window_ev, 0));
     t.evmask |= 1u << 9; /* snappy timed on its own stream: ev9 -> ev8 */
     HIP_TRY(hipEventRecord(t.ev[9], prep));
     hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, prep, d_comp,
@@ -1061,7 +1071,8 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   }
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(5)); /* after copy */
-  HIP_TRY(hipEventRecord(prep_gate_ev, stream)); /* blobs fully consumed */
+  if (window_ev) /* this window's scratch fully consumed */
+    HIP_TRY(hipEventRecord(window_ev, stream));
   hipLaunchKernelGGL(k_rundesc, dim3((ngroups + 255) / 256), dim3(256), 0,
                      stream, groups_for_kernel, ngroups, d_descw, d_totals,
                      d_partial, d_bsums, n, nb, d_place, sl.d_rundescs);
@@ -2045,6 +2056,7 @@ void gra_replay_destroy(GraReplay *r) {
     if (kv.second.d_groups) (void)hipFree(kv.second.d_groups);
     if (kv.second.d_snap) (void)hipFree(kv.second.d_snap);
     if (kv.second.d_ud) (void)hipFree(kv.second.d_ud);
+    if (kv.second.consumed_ev) (void)hipEventDestroy(kv.second.consumed_ev);
   }
   delete r;
 }
@@ -2126,13 +2138,15 @@ int gra_replay_tick(GraReplay *r, uint64_t first, uint64_t n) {
   std::vector<uint32_t> counts(r->counts.begin() + first,
                                r->counts.begin() + first + n);
   std::lock_guard<std::mutex> lk(e->mu);
+  if (r->snappy && !plan.consumed_ev)
+    (void)hipEventCreateWithFlags(&plan.consumed_ev, hipEventDisableTiming);
   return e->enqueue_tick(r->d_blobs, r->d_descs + first, (uint32_t)n,
                          plan.groups, plan.blob_bytes, false, nullptr, 0,
                          nullptr, nullptr, plan.d_groups, r->d_comp,
                          !r->snappy ? nullptr
                          : plan.d_snap ? plan.d_snap
                                        : r->d_snaptasks + first,
-                         std::move(counts));
+                         std::move(counts), plan.consumed_ev);
 }
 
 int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {

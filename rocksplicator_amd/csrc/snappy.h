@@ -38,22 +38,27 @@ WB_HD uint32_t decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
   while (ip < slen) {
     uint8_t tag = src[ip++];
     if ((tag & 3) == 0) { /* literal */
-      /* length math in 64-bit: a 4-extra-byte length near 2^32 must fail
-       * the bounds check, not wrap it (ip+len / op+len in u32 would pass
-       * and the copy loop would scribble ~4 GB past the arena) */
-      uint64_t len64 = (uint32_t)(tag >> 2) + 1;
-      if (len64 > 60) {
-        uint32_t nb = (uint32_t)len64 - 60;
+      uint32_t len = (tag >> 2) + 1;
+      if (__builtin_expect(len > 60, 0)) {
+        /* extended length: math + bounds in 64-bit — a 4-extra-byte length
+         * near 2^32 must FAIL the check, not wrap it (u32 ip+len / op+len
+         * would pass and the copy loop would scribble ~4 GB OOB) */
+        uint32_t nb = len - 60;
         if ((uint64_t)ip + nb > slen) return UINT32_MAX;
-        len64 = 0;
+        uint64_t len64 = 0;
         for (uint32_t b = 0; b < nb; b++)
           len64 |= (uint64_t)src[ip + b] << (8 * b);
         len64 += 1;
         ip += nb;
-      }
-      if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen)
+        if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen)
+          return UINT32_MAX;
+        len = (uint32_t)len64; /* checked: fits the remaining space */
+      } else if (ip + len > slen || op + len > ulen) {
+        /* u32 is exact here: len <= 61 and ip <= slen, op <= ulen, so a
+         * wrap would need slen/ulen > 2^32-62 — i.e. >4 GB of real backing
+         * per slot, within which any access stays in-arena anyway */
         return UINT32_MAX;
-      uint32_t len = (uint32_t)len64; /* checked: fits the remaining space */
+      }
       /* literals never overlap: 16-byte chunks (arena + slot slack cover
        * the over-read/over-write; gfx950 tolerates misaligned dwordx4) */
       for (uint32_t b = 0; b < len; b += 16) {
@@ -84,7 +89,8 @@ WB_HD uint32_t decompress(const uint8_t *src, uint32_t slen, uint8_t *dst,
               ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
         ip += 4;
       }
-      if (off == 0 || off > op || (uint64_t)op + len > ulen) return UINT32_MAX;
+      /* u32 exact: len <= 64 from the tag; wrap needs ulen > 2^32-65 */
+      if (off == 0 || off > op || op + len > ulen) return UINT32_MAX;
       if (off >= 16) { /* chunked forward copy is overlap-safe at off>=chunk */
         for (uint32_t b = 0; b < len; b += 16) {
 #if defined(__HIP_DEVICE_COMPILE__) && defined(WB_UNALIGNED_OK)
@@ -201,20 +207,22 @@ __device__ inline uint32_t decompress_coop16(const uint8_t *__restrict__ src,
   while (ip < slen) {
     uint8_t tag = src[ip++];
     if ((tag & 3) == 0) { /* literal */
-      /* 64-bit length math + bounds, same overflow hazard as decompress() */
-      uint64_t len64 = (uint32_t)(tag >> 2) + 1;
-      if (len64 > 60) {
-        uint32_t nb = (uint32_t)len64 - 60;
+      uint32_t len = (tag >> 2) + 1;
+      if (__builtin_expect(len > 60, 0)) {
+        /* extended length in 64-bit — same overflow hazard as decompress() */
+        uint32_t nb = len - 60;
         if ((uint64_t)ip + nb > slen) return UINT32_MAX;
-        len64 = 0;
+        uint64_t len64 = 0;
         for (uint32_t b = 0; b < nb; b++)
           len64 |= (uint64_t)src[ip + b] << (8 * b);
         len64 += 1;
         ip += nb;
+        if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen)
+          return UINT32_MAX;
+        len = (uint32_t)len64;
+      } else if (ip + len > slen || op + len > ulen) {
+        return UINT32_MAX; /* u32 exact: len <= 61 (see decompress()) */
       }
-      if ((uint64_t)ip + len64 > slen || (uint64_t)op + len64 > ulen)
-        return UINT32_MAX;
-      uint32_t len = (uint32_t)len64;
       for (uint32_t b = lane * 16; b < len; b += 16 * 16)
         *(uint4 *)(dst + op + b) = *(const uint4 *)(src + ip + b);
       ip += len;
@@ -238,7 +246,7 @@ __device__ inline uint32_t decompress_coop16(const uint8_t *__restrict__ src,
               ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
         ip += 4;
       }
-      if (off == 0 || off > op || (uint64_t)op + len > ulen) return UINT32_MAX;
+      if (off == 0 || off > op || op + len > ulen) return UINT32_MAX;
       if (off >= len) { /* no overlap: parallel 16B chunks */
         for (uint32_t b = lane * 16; b < len; b += 16 * 16)
           *(uint4 *)(dst + op + b) = *(const uint4 *)(dst + op - off + b);

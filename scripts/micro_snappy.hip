@@ -487,6 +487,123 @@ __global__ void __launch_bounds__(256) k_v6(const uint8_t *__restrict__ comp,
     snp::decompress(comp + t.comp_off, t.comp_len, out + t.out_off, t.ulen);
 }
 
+/* v7: TWO streams per lane, element-interleaved — the per-element parse is
+ * a dependent byte chain; alternating elements of two independent streams
+ * gives the scheduler a second chain to issue while the first waits. */
+template <int MAXC>
+struct V7St {
+  const uint8_t *src;
+  uint8_t *dst;
+  uint32_t ip, op, slen, ulen;
+  bool active;
+};
+
+template <int MAXC>
+__device__ __forceinline__ void v7_step(V7St<MAXC> &s) {
+  uint8_t tag = s.src[s.ip++];
+  if ((tag & 3) == 0) { /* literal */
+    uint64_t len64 = (uint32_t)(tag >> 2) + 1;
+    if (len64 > 60) {
+      uint32_t nb = (uint32_t)len64 - 60;
+      if ((uint64_t)s.ip + nb > s.slen) { s.active = false; return; }
+      len64 = 0;
+      for (uint32_t b = 0; b < nb; b++)
+        len64 |= (uint64_t)s.src[s.ip + b] << (8 * b);
+      len64 += 1;
+      s.ip += nb;
+    }
+    if ((uint64_t)s.ip + len64 > s.slen || (uint64_t)s.op + len64 > s.ulen) {
+      s.active = false;
+      return;
+    }
+    uint32_t len = (uint32_t)len64;
+    for (uint32_t b = 0; b < len; b += 16)
+      *(uint4 *)(s.dst + s.op + b) = *(const uint4 *)(s.src + s.ip + b);
+    s.ip += len;
+    s.op += len;
+  } else {
+    uint32_t len, off;
+    if ((tag & 3) == 1) {
+      len = ((tag >> 2) & 7) + 4;
+      if (s.ip + 1 > s.slen) { s.active = false; return; }
+      off = ((uint32_t)(tag >> 5) << 8) | s.src[s.ip];
+      s.ip += 1;
+    } else if ((tag & 3) == 2) {
+      len = (tag >> 2) + 1;
+      if (s.ip + 2 > s.slen) { s.active = false; return; }
+      off = (uint32_t)s.src[s.ip] | ((uint32_t)s.src[s.ip + 1] << 8);
+      s.ip += 2;
+    } else {
+      len = (tag >> 2) + 1;
+      if (s.ip + 4 > s.slen) { s.active = false; return; }
+      off = (uint32_t)s.src[s.ip] | ((uint32_t)s.src[s.ip + 1] << 8) |
+            ((uint32_t)s.src[s.ip + 2] << 16) |
+            ((uint32_t)s.src[s.ip + 3] << 24);
+      s.ip += 4;
+    }
+    if (off == 0 || off > s.op || (uint64_t)s.op + len > s.ulen) {
+      s.active = false;
+      return;
+    }
+    if (off >= 16) {
+      for (uint32_t b = 0; b < len; b += 16)
+        *(uint4 *)(s.dst + s.op + b) = *(const uint4 *)(s.dst + s.op - off + b);
+    } else if (off >= 8) {
+      for (uint32_t b = 0; b < len; b += 8)
+        *(uint64_t *)(s.dst + s.op + b) =
+            *(const uint64_t *)(s.dst + s.op - off + b);
+    } else {
+      for (uint32_t b = 0; b < len; b++) s.dst[s.op + b] = s.dst[s.op + b - off];
+    }
+    s.op += len;
+  }
+}
+
+template <int MAXC>
+__global__ void __launch_bounds__(256) k_v7(const uint8_t *__restrict__ comp,
+                                            const Task *__restrict__ tasks,
+                                            uint32_t n,
+                                            uint8_t *__restrict__ out) {
+  constexpr int STRIDE = MAXC + 4;
+  __shared__ uint8_t lds[2 * 256 * STRIDE];
+  uint32_t base2 = (blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  V7St<MAXC> s[2];
+  for (int j = 0; j < 2; j++) {
+    s[j].active = false;
+    uint32_t i = base2 + j;
+    if (i >= n) continue;
+    Task t = tasks[i];
+    if (t.comp_len > MAXC) {
+      snp::decompress(comp + t.comp_off, t.comp_len, out + t.out_off, t.ulen);
+      continue;
+    }
+    uint8_t *mine = lds + ((size_t)threadIdx.x * 2 + j) * STRIDE;
+    const uint8_t *gsrc = comp + t.comp_off;
+    for (uint32_t b = 0; b < t.comp_len; b += 16)
+      *(uint4 *)(mine + b) = *(const uint4 *)(gsrc + b);
+    uint32_t ulen = 0;
+    uint32_t v = wb::varint32(mine, t.comp_len, &ulen);
+    if (v == 0 || ulen > t.ulen) continue;
+    s[j].src = mine;
+    s[j].dst = out + t.out_off;
+    s[j].ip = v;
+    s[j].op = 0;
+    s[j].slen = t.comp_len;
+    s[j].ulen = ulen;
+    s[j].active = true;
+  }
+  while (s[0].active || s[1].active) {
+    if (s[0].active) {
+      v7_step<MAXC>(s[0]);
+      if (s[0].ip >= s[0].slen) s[0].active = false;
+    }
+    if (s[1].active) {
+      v7_step<MAXC>(s[1]);
+      if (s[1].ip >= s[1].slen) s[1].active = false;
+    }
+  }
+}
+
 int main(int argc, char **argv) {
   uint32_t n = argc > 1 ? atoi(argv[1]) : 400000;
   uint32_t vlen = argc > 2 ? atoi(argv[2]) : 1024;
@@ -662,6 +779,48 @@ int main(int argc, char **argv) {
     run("v6c4-252-sort", [&] {
       hipLaunchKernelGGL((k_v6<4, 252>), dim3(((size_t)n * 4 + 255) / 256),
                          dim3(256), 0, 0, d_comp, d_sorted, n, d_out);
+    });
+    /* split-launch: the sorted prefix that fits a small LDS stage runs in
+     * a high-occupancy kernel (32/64 KB LDS -> 4-5/2-3 blocks per CU), the
+     * tail keeps the 128 KB stage. No global-parse fallback anywhere. */
+    uint32_t p124 = 0, p252 = 0;
+    while (p124 < n && sorted[p124].comp_len <= 124) p124++;
+    while (p252 < n && sorted[p252].comp_len <= 252) p252++;
+    printf("split points: <=124: %u (%.0f%%)  <=252: %u (%.0f%%)\n", p124,
+           100.0 * p124 / n, p252, 100.0 * p252 / n);
+    run("split252/508", [&] {
+      if (p252)
+        hipLaunchKernelGGL(k_v2<252>, dim3((p252 + 255) / 256), dim3(256), 0,
+                           0, d_comp, d_sorted, p252, d_out);
+      if (n - p252)
+        hipLaunchKernelGGL(k_v2<508>, dim3((n - p252 + 255) / 256), dim3(256),
+                           0, 0, d_comp, d_sorted + p252, n - p252, d_out);
+    });
+    run("split124/252/508", [&] {
+      if (p124)
+        hipLaunchKernelGGL(k_v2<124>, dim3((p124 + 255) / 256), dim3(256), 0,
+                           0, d_comp, d_sorted, p124, d_out);
+      if (p252 - p124)
+        hipLaunchKernelGGL(k_v2<252>, dim3((p252 - p124 + 255) / 256),
+                           dim3(256), 0, 0, d_comp, d_sorted + p124,
+                           p252 - p124, d_out);
+      if (n - p252)
+        hipLaunchKernelGGL(k_v2<508>, dim3((n - p252 + 255) / 256), dim3(256),
+                           0, 0, d_comp, d_sorted + p252, n - p252, d_out);
+    });
+    run("v7-ilv2-252", [&] {
+      hipLaunchKernelGGL(k_v7<252>, dim3((n / 2 + 255) / 256), dim3(256), 0,
+                         0, d_comp, d_sorted, n, d_out);
+    });
+    run("split188/508", [&] {
+      uint32_t p188 = 0;
+      while (p188 < n && sorted[p188].comp_len <= 188) p188++;
+      if (p188)
+        hipLaunchKernelGGL(k_v2<188>, dim3((p188 + 255) / 256), dim3(256), 0,
+                           0, d_comp, d_sorted, p188, d_out);
+      if (n - p188)
+        hipLaunchKernelGGL(k_v2<508>, dim3((n - p188 + 255) / 256), dim3(256),
+                           0, 0, d_comp, d_sorted + p188, n - p188, d_out);
     });
     CHECK(hipFree(d_sorted));
   }
