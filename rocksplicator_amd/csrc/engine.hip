@@ -1030,21 +1030,30 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
     HIP_TRY(hipStreamWaitEvent(stream, t.ev[1], 0)); /* kernels gate on data */
   }
   uint32_t nb = (n + 255) / 256;
+  static const bool snappy_overlap = [] { /* A/B escape hatch */
+    const char *v = getenv("GRA_SNAPPY_OVERLAP");
+    return !v || v[0] != '0';
+  }();
   if (d_snaptasks) { /* config #5 pre-stage: decompress into the blob arena
                       * on the prep stream, overlapped with the previous
                       * tick's decode..copy. window_ev (recorded after the
                       * LAST tick of this same window finished reading the
                       * scratch) fences window reuse without serializing
-                      * prep behind main. */
-    if (window_ev) HIP_TRY(hipStreamWaitEvent(prep, window_ev, 0));
+                      * prep behind main. Both kernels are CU-bound, so the
+                      * overlap mostly timeslices — kept because it never
+                      * loses and wins when the main tick has SDMA phases. */
+    hipStream_t ps = snappy_overlap ? prep : stream;
+    if (window_ev && snappy_overlap)
+      HIP_TRY(hipStreamWaitEvent(prep, window_ev, 0));
     t.evmask |= 1u << 9; /* snappy timed on its own stream: ev9 -> ev8 */
-    HIP_TRY(hipEventRecord(t.ev[9], prep));
-    hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, prep, d_comp,
+    HIP_TRY(hipEventRecord(t.ev[9], ps));
+    hipLaunchKernelGGL(k_snappy, dim3(nb), dim3(256), 0, ps, d_comp,
                        d_snaptasks, n, (uint8_t *)d_blobs, d_err_ring, tick);
     HIP_TRY(hipGetLastError());
     t.evmask |= 1u << 8;
-    HIP_TRY(hipEventRecord(t.ev[8], prep));
-    HIP_TRY(hipStreamWaitEvent(stream, t.ev[8], 0)); /* decode gates on it */
+    HIP_TRY(hipEventRecord(t.ev[8], ps));
+    if (snappy_overlap)
+      HIP_TRY(hipStreamWaitEvent(stream, t.ev[8], 0)); /* decode gates on it */
   }
   hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
                      n, d_totals, opts.max_wb_records, d_err_ring, tick,
