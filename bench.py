@@ -109,6 +109,56 @@ def h2d_side_leg(eng, arena, used, descs, n_upd, args, steps=24, warmup=6):
     }
 
 
+def drain_side_leg(arena_p, descs, n_upd, args, steps=8, warmup=3):
+    """the host-drain leg — apply into the device store ring
+    AND stream every tick's headers+payload into pinned host arenas
+    (k_drain on the copyout stream, overlapped with the next tick's
+    kernels): the north star's literal 'drain sorted runs back to the host
+    RocksDB memtables'. Reported beside the PCIe D2H bound."""
+    ticks = max(1, min(2, n_upd // args.tick_updates))
+    n = ticks * args.tick_updates
+    end = descs[n - 1].off + descs[n - 1].len
+    per_tick = end // ticks
+    eng = ra.Engine(nshards=args.nshards, device=-1, store_ring=1,
+                    drain_host=1,
+                    store_bytes=min(per_tick * 14 + (1 << 30), 48 << 30))
+    lib = ra.load()
+    arena_need = per_tick + 48 * args.tick_updates + 4096
+    rc = lib.gra_drain_prewarm(eng.h, arena_need, steps + warmup + 2)
+    assert rc == 0, ra.ffi.last_error(lib)
+    rep = eng.upload(arena_p, end, descs, n)
+    for w in range(ticks):
+        rep.prepare(w * args.tick_updates, args.tick_updates)
+
+    def step(i):
+        rep.tick((i % ticks) * args.tick_updates, args.tick_updates)
+
+    for i in range(warmup):
+        step(i)
+    rep.sync()
+    eng.stats_reset()
+    t0 = time.perf_counter()
+    for i in range(steps):
+        step(warmup + i)
+    rep.sync()
+    wall = time.perf_counter() - t0
+    st = eng.stats()
+    d2h_bytes = st.payload_bytes + 24 * st.records  # arena bytes drained
+    out = {
+        "value": steps * args.tick_updates / wall,
+        "unit": "updates/s",
+        "ms_per_step": wall * 1e3 / steps,
+        "d2h_gbps": d2h_bytes / wall / 1e9,
+        "steps": steps,
+        "note": "drain-host side leg: apply + k_drain of every tick's "
+                "hdrs+payload into pinned host arenas on the copyout "
+                "stream; host runs own the bytes (ring store recycles "
+                "device side)",
+    }
+    eng.close()
+    return out
+
+
 def cpu_baseline_leg(raw_arena, used, descs, n, nshards, target_s, snappy=False):
     """Time the ORACLE applier (the CPU restatement of replicated_db.cpp:
     369-383 + rocksdb_wrapper.cpp:13-28) on the host cores — checker/baseline
@@ -439,6 +489,10 @@ def main():
     if rank == 0 and world == 1 and not args.h2d and not args.snappy:
         # staging-inclusive side line on every headline run
         result["h2d_staged"] = h2d_side_leg(eng, arena, used, descs, n_upd, args)
+        # host-drain side line; separate engine so the
+        # drain mode never touches the headline engine's state
+        result["drain_host"] = drain_side_leg(
+            C.cast(arena, C.POINTER(C.c_uint8)), descs, n_upd, args)
 
     if rank == 0 and world == 1 and args.cpu_baseline:  # contract: N=1 only
         if args.snappy:

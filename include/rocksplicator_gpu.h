@@ -114,6 +114,10 @@ int gra_get(GraDb *db, const void *key, size_t klen, void *buf, size_t cap,
  * this, gra_latest_seq and gra_get are linearizable per shard. */
 int gra_flush(GraEngine *e);
 
+/* drain_host mode setup: pre-pin `count` host arenas of `bytes` each so no
+ * hipHostMalloc happens inside a timed region. */
+int gra_drain_prewarm(GraEngine *e, size_t bytes, uint32_t count);
+
 /* Batched point reads served FROM THE DEVICE STORE (followers serve reads
  * in rocksplicator deployments — ApplicationDB::Get routes to the local db;
  * here the memtable lives in HBM, so the search runs there too: one block

@@ -394,6 +394,30 @@ __global__ void __launch_bounds__(256) k_copy(const uint8_t *__restrict__ blobs,
   }
 }
 
+/* Drain-host mode (the north star's literal "drain sorted runs back to the
+ * host memtables"): stream the tick's header + payload regions from the
+ * device store STRAIGHT INTO mapped pinned host memory. A kernel (not
+ * hipMemcpyAsync) because only the device knows the tick's placement
+ * (TickPlace is computed by k_scan2); grid-stride uint4 writes over PCIe.
+ * Host arena layout: [hdrs, 16-aligned][payload]. */
+__global__ void k_drain(const uint8_t *__restrict__ store,
+                        const TickPlace *__restrict__ place,
+                        uint8_t *__restrict__ hostbuf) {
+  if (place->overflow) return;
+  uint64_t hdr_bytes = (uint64_t)place->total_rec * sizeof(wb::RecHdr);
+  uint64_t hb16 = (hdr_bytes + 15) & ~15ull;
+  uint64_t pay = place->payload_bytes;
+  const uint8_t *hsrc = store + place->hdr_off;
+  const uint8_t *psrc = store + place->payload_off;
+  size_t stride = (size_t)gridDim.x * blockDim.x * 16;
+  size_t i0 = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 16;
+  for (size_t b = i0; b < hdr_bytes; b += stride)
+    *(uint4 *)(hostbuf + b) = *(const uint4 *)(hsrc + b);
+  uint8_t *pdst = hostbuf + hb16;
+  for (size_t b = i0; b < pay; b += stride)
+    *(uint4 *)(pdst + b) = *(const uint4 *)(psrc + b);
+}
+
 __global__ void k_rundesc(const GroupDesc *__restrict__ groups, uint32_t ngroups,
                           const UpdDesc *__restrict__ descs,
                           const wb::WalkTotals *__restrict__ totals,
@@ -613,11 +637,13 @@ struct TickRec {
   bool h2d_timed = false;
   uint32_t evmask = 0; /* which ev[i] were recorded this tick */
   std::vector<uint32_t> counts; /* per-update record counts (error recovery) */
+  std::shared_ptr<uint8_t> drain_buf; /* drain-host: this tick's pinned arena */
   hipEvent_t ev[kEventsPerTick];
 };
 
 struct Slot {
   GroupDesc *h_groups = nullptr;   /* pinned */
+  TickPlace *h_place = nullptr;    /* pinned mirror of this tick's placement */
   DevRunDesc *h_rundescs = nullptr;
   DevRunDesc *d_rundescs = nullptr; /* per-slot device buffer so the D2H can
                                        overlap the next tick's kernels */
@@ -665,7 +691,9 @@ struct GraEngine {
   uint2 *d_partial = nullptr, *d_bsums = nullptr;
   wb::Rec *d_reccache = nullptr;
   CopyTask *d_tasks = nullptr;
-  TickPlace *d_place = nullptr;
+  TickPlace *d_place = nullptr; /* kSlots entries: per-tick placement slot
+                                   (the drain D2H snapshot on copyout must
+                                   not race the next tick's k_scan2) */
   GroupDesc *d_groups = nullptr;
   uint32_t *d_err_ring = nullptr; /* kErrRing slots, zeroed at init */
   uint32_t tick_id = 0;
@@ -716,6 +744,13 @@ struct GraEngine {
     GraGetResult *d_out = nullptr;
     size_t runs_cap = 0, keys_cap = 0, keybuf_cap = 0, valbuf_cap = 0, out_cap = 0;
   } mg;
+  /* drain-host pinned arena pool. DECLARED BEFORE shards: runs hold
+   * shared_ptrs whose deleter returns buffers here, so the pool must be
+   * destroyed after the shards release them. */
+  std::mutex drain_mu;
+  std::vector<std::pair<uint8_t *, size_t>> drain_pool;
+  bool closing = false; /* deleter frees directly during teardown */
+  std::shared_ptr<uint8_t> drain_alloc(size_t need);
   /* shards */
   std::vector<ShardState> shards;
   Stats stats;
@@ -786,6 +821,35 @@ struct GraReplay {
   std::map<std::pair<uint64_t, uint64_t>, TickPlan> plans; /* window cache */
 };
 
+std::shared_ptr<uint8_t> GraEngine::drain_alloc(size_t need) {
+  uint8_t *p = nullptr;
+  size_t cap = 0;
+  {
+    std::lock_guard<std::mutex> lk(drain_mu);
+    for (auto it = drain_pool.begin(); it != drain_pool.end(); ++it) {
+      if (it->second >= need) {
+        p = it->first;
+        cap = it->second;
+        drain_pool.erase(it);
+        break;
+      }
+    }
+  }
+  if (!p) {
+    cap = need + 64;
+    if (hipHostMalloc(&p, cap) != hipSuccess) return nullptr;
+  }
+  GraEngine *e = this;
+  return std::shared_ptr<uint8_t>(p, [e, cap](uint8_t *q) {
+    std::lock_guard<std::mutex> lk(e->drain_mu);
+    if (e->closing) {
+      (void)hipHostFree(q);
+    } else {
+      e->drain_pool.push_back({q, cap});
+    }
+  });
+}
+
 hipEvent_t GraEngine::get_event() {
   if (!event_pool.empty()) {
     hipEvent_t e = event_pool.back();
@@ -836,7 +900,7 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMalloc(&d_bsums, ((size_t)max_upd / 256 + 2) * sizeof(uint2)));
   HIP_TRY(hipMalloc(&d_reccache, (size_t)max_upd * 2 * sizeof(wb::Rec)));
   HIP_TRY(hipMalloc(&d_tasks, (size_t)task_cap * sizeof(CopyTask)));
-  HIP_TRY(hipMalloc(&d_place, sizeof(TickPlace)));
+  HIP_TRY(hipMalloc(&d_place, kSlots * sizeof(TickPlace)));
   HIP_TRY(hipMalloc(&d_groups, (size_t)group_cap * sizeof(GroupDesc)));
   HIP_TRY(hipMalloc(&d_err_ring, kErrRing * 4));
   HIP_TRY(hipMemset(d_err_ring, 0, kErrRing * 4));
@@ -865,6 +929,7 @@ int GraEngine::init(const GraEngineOpts &o) {
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
     HIP_TRY(hipHostMalloc(&s.h_groups, (size_t)group_cap * sizeof(GroupDesc)));
+    HIP_TRY(hipHostMalloc(&s.h_place, sizeof(TickPlace)));
     HIP_TRY(hipHostMalloc(&s.h_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
     HIP_TRY(hipMalloc(&s.d_rundescs, (size_t)group_cap * sizeof(DevRunDesc)));
     HIP_TRY(hipMalloc(&s.d_ok, (size_t)max_upd));
@@ -885,6 +950,10 @@ int GraEngine::init(const GraEngineOpts &o) {
 }
 
 GraEngine::~GraEngine() {
+  {
+    std::lock_guard<std::mutex> lk(drain_mu);
+    closing = true; /* run releases during/after teardown free directly */
+  }
   (void)hipStreamSynchronize(stream);
   (void)hipStreamSynchronize(copyout); /* pending rundesc/ok D2H target the
                                           pinned slot buffers freed below */
@@ -901,6 +970,7 @@ GraEngine::~GraEngine() {
   for (int i = 0; i < kSlots; i++) {
     Slot &s = slots[i];
     if (s.h_groups) (void)hipHostFree(s.h_groups);
+    if (s.h_place) (void)hipHostFree(s.h_place);
     if (s.h_rundescs) (void)hipHostFree(s.h_rundescs);
     if (s.d_rundescs) (void)hipFree(s.d_rundescs);
     if (s.d_ok) (void)hipFree(s.d_ok);
@@ -922,6 +992,13 @@ GraEngine::~GraEngine() {
                   (void *)d_tasks, (void *)d_place, (void *)d_groups,
                   (void *)d_err_ring})
     if (p) (void)hipFree(p);
+  shards.clear(); /* release run arenas before draining the pool */
+  pending.clear();
+  {
+    std::lock_guard<std::mutex> lk(drain_mu);
+    for (auto &pr : drain_pool) (void)hipHostFree(pr.first);
+    drain_pool.clear();
+  }
   if (stream) (void)hipStreamDestroy(stream);
   if (copyout) (void)hipStreamDestroy(copyout);
   if (h2d) (void)hipStreamDestroy(h2d);
@@ -1060,23 +1137,24 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                      d_partial, d_bsums, d_reccache, sl.d_ok);
   HIP_TRY(hipGetLastError());
   if (detailed) HIP_TRY(rec(2)); /* after decode(+scan1) */
+  TickPlace *place_slot = d_place + si;
   hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, d_bsums, nb,
-                     d_cursor, d_place, opts.store_bytes, opts.store_ring,
+                     d_cursor, place_slot, opts.store_bytes, opts.store_ring,
                      task_cap, d_err_ring, tick);
   HIP_TRY(hipGetLastError());
   if (detailed) HIP_TRY(rec(3)); /* after scan(+reserve) */
   hipLaunchKernelGGL(k_emit, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
-                     n, d_totals, d_partial, d_bsums, d_place, d_store,
+                     n, d_totals, d_partial, d_bsums, place_slot, d_store,
                      d_tasks, d_reccache);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(4)); /* after emit (pre-copy) */
   /* group width by average update size (micro_copy.hip: g32 wins >=512B) */
   if (blob_bytes / (n ? n : 1) >= 512) {
     hipLaunchKernelGGL((k_copy<32>), dim3(2048), dim3(256), 0, stream, d_blobs,
-                       d_store, d_place, d_tasks);
+                       d_store, place_slot, d_tasks);
   } else {
     hipLaunchKernelGGL((k_copy<16>), dim3(2048), dim3(256), 0, stream, d_blobs,
-                       d_store, d_place, d_tasks);
+                       d_store, place_slot, d_tasks);
   }
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(5)); /* after copy */
@@ -1084,7 +1162,7 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
     HIP_TRY(hipEventRecord(window_ev, stream));
   hipLaunchKernelGGL(k_rundesc, dim3((ngroups + 255) / 256), dim3(256), 0,
                      stream, groups_for_kernel, ngroups, d_descw, d_totals,
-                     d_partial, d_bsums, n, nb, d_place, sl.d_rundescs);
+                     d_partial, d_bsums, n, nb, place_slot, sl.d_rundescs);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(6)); /* main-stream tick end */
   if (stage_buf >= 0) /* last reader of this staging buffer has retired */
@@ -1099,6 +1177,24 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   HIP_TRY(hipMemcpyAsync(sl.h_err, d_err_ring + (tick % kErrRing), 4,
                          hipMemcpyDeviceToHost, copyout));
   HIP_TRY(hipMemcpyAsync(sl.h_ok, sl.d_ok, n, hipMemcpyDeviceToHost, copyout));
+  HIP_TRY(hipMemcpyAsync(sl.h_place, place_slot, sizeof(TickPlace),
+                         hipMemcpyDeviceToHost, copyout));
+  if (opts.drain_host && !t.counts.empty()) {
+    /* drain the whole tick into a pinned host arena on the copyout stream,
+     * overlapped with the next tick's kernels. Size bound: payload <=
+     * blob + 7/record (cf re-prefix), + 15/record align pad + 24/record
+     * headers. Without counts (no record total) ingest falls back to the
+     * eager per-run fetch. */
+    uint64_t recs = 0;
+    for (uint32_t c : t.counts) recs += c;
+    size_t need = (size_t)blob_bytes + 46 * recs + 80;
+    t.drain_buf = drain_alloc(need);
+    if (t.drain_buf) {
+      hipLaunchKernelGGL(k_drain, dim3(1024), dim3(256), 0, copyout, d_store,
+                         place_slot, t.drain_buf.get());
+      HIP_TRY(hipGetLastError());
+    }
+  }
   HIP_TRY(hipEventRecord(t.ev[7], copyout)); /* publication done */
   pending.push_back(std::move(t));
   return GRA_OK;
@@ -1156,6 +1252,19 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
   stats.updates += t.n;
   stats.blob_bytes += t.blob_bytes;
   uint32_t err = *sl.h_err;
+  /* drain-host: back a run with its span of the tick's pinned arena
+   * (k_drain streamed the whole tick there; kv_off stays tick-relative,
+   * pay_p = tick payload base). Falls back to the eager per-run fetch when
+   * the arena is absent (no counts / allocation failure). */
+  auto attach_drain = [&](Run &run, const DevRunDesc &rd) -> bool {
+    if (!t.drain_buf || sl.h_place->overflow) return false;
+    uint64_t hb16 =
+        ((uint64_t)sl.h_place->total_rec * sizeof(wb::RecHdr) + 15) & ~15ull;
+    run.hbuf = t.drain_buf;
+    run.hdr_p = t.drain_buf.get() + (rd.hdr_off - sl.h_place->hdr_off);
+    run.pay_p = t.drain_buf.get() + hb16;
+    return true;
+  };
   if (err != 0) {
     /* Precise corruption recovery: per-update validity came back with the
      * run descriptors (sl.h_ok). Per shard, keep the prefix of records from
@@ -1219,7 +1328,8 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
         run->hdr_cur = rd.hdr_off;
         run->payload_cur = rd.payload_off;
         run->pay_rel_base = rd.pay_rel_base;
-        if (opts.drain_host) (void)fetch_run_impl(this, *run);
+        if (opts.drain_host && !attach_drain(*run, rd))
+          (void)fetch_run_impl(this, *run);
         if (rd.n_entries > 0 && !opts.store_ring) ss.runs.push_back(std::move(run));
         if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
         stats.records += rd.n_entries;
@@ -1236,7 +1346,8 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
         run->hdr_cur = rd.hdr_off;
         run->payload_cur = rd.payload_off;
         run->pay_rel_base = rd.pay_rel_base;
-        if (opts.drain_host) (void)fetch_run_impl(this, *run);
+        if (opts.drain_host && !attach_drain(*run, rd))
+          (void)fetch_run_impl(this, *run);
         ss.runs.push_back(std::move(run));
         stats.records += keep_recs;
       }
@@ -1283,6 +1394,18 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
         drop_stale_locked(ss);
         continue;
       }
+      if (opts.drain_host && rd.n_entries > 0) {
+        /* ring + drain = the production drain shape: the device store is a
+         * staging ring, the host arena keeps the durable run (host-only:
+         * device bytes recycle, so reads route through the host path) */
+        auto run = std::make_shared<Run>();
+        run->base_seq = rd.base_seq;
+        run->last_seq = rd.last_seq;
+        run->n_entries = rd.n_entries;
+        run->payload_bytes = rd.payload_bytes;
+        run->pay_rel_base = rd.pay_rel_base;
+        if (attach_drain(*run, rd)) ss.runs.push_back(std::move(run));
+      }
       if (rd.last_seq > ss.durable_seq) ss.durable_seq = rd.last_seq;
       stats.records += rd.n_entries;
       stats.payload_bytes += rd.payload_bytes;
@@ -1308,7 +1431,8 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
     run->payload_cur = rd.payload_off;
     run->pay_rel_base = rd.pay_rel_base;
     if (rd.n_entries > 0) {
-      if (opts.drain_host) (void)fetch_run_impl(this, *run); /* eager D2H */
+      if (opts.drain_host && !attach_drain(*run, rd))
+        (void)fetch_run_impl(this, *run); /* eager D2H fallback */
       ss.runs.push_back(std::move(run));
     }
     ss.durable_seq = rd.last_seq > ss.durable_seq ? rd.last_seq : ss.durable_seq;
@@ -1722,6 +1846,22 @@ int gra_flush(GraEngine *e) {
   return e->flush_locked();
 }
 
+/* Pre-pin `count` drain arenas of `bytes` each into the pool (harness
+ * setup: pinning ~1 GB costs ~100 ms — keep it out of timed regions). */
+int gra_drain_prewarm(GraEngine *e, size_t bytes, uint32_t count) {
+  std::vector<std::shared_ptr<uint8_t>> held;
+  held.reserve(count);
+  for (uint32_t i = 0; i < count; i++) {
+    auto p = e->drain_alloc(bytes);
+    if (!p) {
+      g_err = "gra_drain_prewarm: pinned allocation failed";
+      return GRA_ERR;
+    }
+    held.push_back(std::move(p));
+  }
+  return GRA_OK; /* releasing `held` returns every arena to the pool */
+}
+
 static int fetch_run_impl(GraEngine *e, Run &r) {
   if (r.resident()) return GRA_OK;
   r.hdrs.resize((size_t)r.n_entries * sizeof(wb::RecHdr));
@@ -1855,12 +1995,13 @@ int gra_shard_checksum(GraDb *db, uint64_t *out) {
     for (const auto &rp : ss.runs) {
       const Run &r = *rp;
       if (r.n_entries == 0) continue;
-      if (r.hdr_cur == UINT64_MAX) { /* host-origin: fold on the host */
-        const wb::RecHdr *h = (const wb::RecHdr *)r.hdrs.data();
+      if (r.hdr_cur == UINT64_MAX) { /* host-origin/drained: fold on host */
+        const wb::RecHdr *h = (const wb::RecHdr *)r.hdrs_data();
+        const uint8_t *pay = r.payload_data();
         for (uint32_t i = 0; i < r.n_entries; i++)
           host_sum += rec_hash_cs(h[i].seq, h[i].type, h[i].key_len,
-                                  h[i].val_len, r.payload.data() + h[i].kv_off,
-                                  r.payload.data() + h[i].kv_off + h[i].key_len);
+                                  h[i].val_len, pay + h[i].kv_off,
+                                  pay + h[i].kv_off + h[i].key_len);
       } else {
         views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
       }

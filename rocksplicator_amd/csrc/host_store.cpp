@@ -32,8 +32,8 @@ int run_get(const std::vector<std::shared_ptr<Run>> &runs, const void *key_,
   for (auto it = runs.rbegin(); it != runs.rend() && !stopped; ++it) {
     const Run &r = **it;
     if (r.n_entries == 0) continue;
-    const wb::RecHdr *hdrs = (const wb::RecHdr *)r.hdrs.data();
-    const uint8_t *pay = r.payload.data();
+    const wb::RecHdr *hdrs = (const wb::RecHdr *)r.hdrs_data();
+    const uint8_t *pay = r.payload_data();
     for (int32_t i = (int32_t)r.n_entries - 1; i >= 0; i--) {
       const wb::RecHdr &h = hdrs[i];
       if (h.type == wb::kRangeDeletion) {

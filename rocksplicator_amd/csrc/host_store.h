@@ -38,7 +38,16 @@ struct Run {
   /* host copies — always set for host runs, lazily fetched for device runs */
   std::vector<uint8_t> hdrs;     /* n_entries × wb::RecHdr */
   std::vector<uint8_t> payload;
-  bool resident() const { return !hdrs.empty() || n_entries == 0; }
+  /* drain-host mode: runs are SPANS of a tick-shared pinned arena drained
+   * by k_drain (no per-run host copy). kv_off stays tick-relative; pay_p
+   * points at the tick's payload base so pay_p + kv_off is correct. */
+  std::shared_ptr<uint8_t> hbuf;
+  const uint8_t *hdr_p = nullptr, *pay_p = nullptr;
+  const uint8_t *hdrs_data() const { return hdr_p ? hdr_p : hdrs.data(); }
+  const uint8_t *payload_data() const { return pay_p ? pay_p : payload.data(); }
+  bool resident() const {
+    return hdr_p != nullptr || !hdrs.empty() || n_entries == 0;
+  }
 };
 
 /* One retained batch for downstream serving (leader update log — the WAL

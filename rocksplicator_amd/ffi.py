@@ -142,6 +142,7 @@ def load():
     lib.gra_replay_destroy.argtypes = [C.c_void_p]
     lib.gra_replay_tick.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
     lib.gra_replay_prepare.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
+    lib.gra_drain_prewarm.argtypes = [C.c_void_p, C.c_size_t, C.c_uint32]
     lib.gra_replay_tick_h2d.argtypes = [C.c_void_p, C.c_uint64, C.c_uint64]
     lib.gra_replay_sync.argtypes = [C.c_void_p]
     lib.gra_stats.argtypes = [C.c_void_p, C.POINTER(GraStats)]

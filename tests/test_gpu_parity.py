@@ -949,3 +949,61 @@ def test_cf_range_delete_parity(olib):
     # bit-level store parity
     assert db.checksum() == olib_checksum(olib, ost)
     e.close()
+
+
+def test_drain_ring_mode_parity(olib):
+    """Ring store + drain_host: the production drain shape — device store
+    recycles, pinned host arenas own the runs (k_drain path). Get, checksum
+    and seq must equal the oracle; multiget routes through the host runs."""
+    nshards, nupd = 16, 8000
+    arena, used, descs = ra.gen_stream(nshards=nshards, n_updates=nupd,
+                                       key_len=16, val_len=96, kind=2,
+                                       seed=77)
+    raw = bytes(arena)[:used]
+    ost = oracle_apply_stream(olib, nshards, raw, descs, nupd, merge_op=1)
+    e = ra.Engine(nshards=nshards, store_ring=1, drain_host=1, merge_op=1)
+    rep = e.upload(C.cast(arena, C.POINTER(C.c_uint8)), used, descs, nupd)
+    for first in range(0, nupd, 2000):  # several ticks -> several arenas
+        rep.tick(first, 2000)
+    rep.sync()
+    keys = collect_keys(olib, raw, descs, 4000)
+    check_parity(e, ost, keys, range(nshards))
+    for s in range(0, nshards, 4):
+        db = e.open(s)
+        assert db.checksum() == olib_checksum(olib, ost, s), s
+        ks = keys.get(s, [])[:16]
+        if ks:
+            for k, v in zip(ks, db.multiget(ks)):
+                assert v == ost.get(s, k), (s, k)
+        db.close()
+    e.close()
+
+
+def test_drain_streaming_with_corruption(olib):
+    """Drained runs + the corruption/rollback machinery interact correctly:
+    a corrupt batch mid-stream must not leave phantom drained content."""
+    e = ra.Engine(nshards=1, store_ring=1, drain_host=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1)
+    good = [PyBatch().put(f"k{i}".encode(), f"v{i}".encode()).data()
+            for i in range(6)]
+    bad = bytearray(PyBatch().put(b"x", b"y").data())
+    bad[8] = 5
+    for rep in good[:3]:
+        assert db.handle_replicate_response(rep)
+        assert ost.apply(0, rep)
+    e.flush()
+    assert db.handle_replicate_response(bytes(bad))
+    assert not ost.apply(0, bytes(bad))
+    e.flush()
+    assert not db.handle_replicate_response(good[3])  # fail-once
+    for rep in good[3:]:
+        assert db.handle_replicate_response(rep)
+        assert ost.apply(0, rep)
+    e.flush()
+    assert db.latest_seq() == ost.latest_seq(0) == 6
+    for i in range(6):
+        assert db.get(f"k{i}".encode()) == f"v{i}".encode()
+    assert db.get(b"x") is None
+    assert db.checksum() == olib_checksum(olib, ost)
+    e.close()
