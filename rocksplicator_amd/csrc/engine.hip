@@ -283,9 +283,11 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = btag;
     h.flags = cf4 ? 1 : 0;
-    h.kpref = 0; /* reserved; a key-prefix filter here measured NEGATIVE:
-                    emit would re-touch blob key bytes (+35us/tick cold)
-                    for no multiget win */
+    /* kpref = first 4 STORED key bytes (cf records: the cf id itself —
+     * free here; plain records: k_copy fills it from the bytes it is
+     * copying anyway). Lets k_multiget reject non-matching entries on the
+     * header alone, no payload touch. */
+    h.kpref = cf4 ? r.cf_id : 0;
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -293,7 +295,9 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     CopyTask tk;
     tk.src_off = d.off + r.key_off;
     tk.dst_rel = pay + cf4;
-    tk.nbytes = r.key_len;
+    /* bit31 marks a cf key task: k_copy must NOT overwrite the kpref emit
+     * just set (the stored prefix is the cf id, not the raw key bytes) */
+    tk.nbytes = r.key_len | (cf4 ? 0x80000000u : 0u);
     tasks[2 * (rec + idx)] = tk;
     if (cfv) { /* unaligned position: byte stores */
       uint8_t *p = pay_region + pay + cf4 + r.key_len;
@@ -387,10 +391,18 @@ __global__ void __launch_bounds__(256) k_copy(const uint8_t *__restrict__ blobs,
   uint32_t g = (blockIdx.x * blockDim.x + threadIdx.x) / G;
   uint32_t ngroups = (gridDim.x * blockDim.x) / G;
   uint8_t *pay_region = store + place->payload_off;
+  wb::RecHdr *hdrs = (wb::RecHdr *)(store + place->hdr_off);
   for (uint32_t t = g; t < ntasks; t += ngroups) {
     CopyTask tk = tasks[t];
-    if (tk.nbytes == 0) continue;
-    copy_dwordx4<G>(pay_region + tk.dst_rel, blobs + tk.src_off, tk.nbytes, lane);
+    uint32_t nb = tk.nbytes & 0x7FFFFFFFu;
+    if ((t & 1) == 0 && lane == 0 && !(tk.nbytes >> 31)) {
+      /* plain-key task: fill the header's key prefix from the bytes this
+       * group is about to stream anyway (same cachelines — near free; the
+       * r01 emit-side attempt re-touched COLD blob bytes and lost) */
+      hdrs[t >> 1].kpref = wb::key_prefix4(blobs + tk.src_off, nb);
+    }
+    if (nb == 0) continue;
+    copy_dwordx4<G>(pay_region + tk.dst_rel, blobs + tk.src_off, nb, lane);
   }
 }
 
@@ -470,12 +482,20 @@ __device__ inline int dev_memcmp(const uint8_t *a, const uint8_t *b, uint32_t n)
   return 0;
 }
 
+/* per-query raw seq info so mixed device/host shards can merge the device
+ * verdict with a host-run probe (gra_multiget) */
+struct MgExtra {
+  uint64_t term_seq, merge_seq, rd_seq;
+  uint32_t term_type, _pad;
+};
+
 __global__ void k_multiget(const uint8_t *__restrict__ store,
                            const RunView *__restrict__ runs, uint32_t nruns,
                            const GraKeyRef *__restrict__ keys,
                            const uint8_t *__restrict__ keybuf, uint32_t nq,
                            uint8_t *__restrict__ valbuf, uint32_t val_stride,
-                           GraGetResult *__restrict__ out) {
+                           GraGetResult *__restrict__ out,
+                           MgExtra *__restrict__ extra) {
   __shared__ uint64_t sh_term_seq[256];
   __shared__ uint64_t sh_term_ref[256]; /* (run<<32)|entry, ~0 = none */
   __shared__ uint64_t sh_merge_seq[256];
@@ -484,6 +504,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   if (q >= nq) return;
   const uint8_t *key = keybuf + keys[q].off;
   uint32_t klen = keys[q].len;
+  uint32_t qpref = wb::key_prefix4(key, klen);
   uint64_t term_seq = 0, term_ref = ~0ULL, merge_seq = 0, rd_seq = 0;
   for (uint32_t r = 0; r < nruns; r++) {
     RunView rv = runs[r];
@@ -491,6 +512,9 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
     const uint8_t *pay = store + rv.payload_off;
     for (uint32_t i = threadIdx.x; i < rv.n_entries; i += blockDim.x) {
       wb::RecHdr h = hdrs[i];
+      if (h.type != wb::kRangeDeletion &&
+          (h.key_len != klen || h.kpref != qpref))
+        continue; /* header-only reject: no payload touch */
       uint32_t rel = h.kv_off - rv.pay_rel_base;
       if (h.type == wb::kRangeDeletion) {
         const uint8_t *b = pay + rel, *e2 = pay + rel + h.key_len;
@@ -502,7 +526,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
         if (h.seq > rd_seq) rd_seq = h.seq;
         continue;
       }
-      if (h.key_len != klen || dev_memcmp(pay + rel, key, klen) != 0)
+      if (dev_memcmp(pay + rel, key, klen) != 0) /* len+prefix pre-checked */
         continue;
       if (h.type == wb::kMerge) {
         if (h.seq > merge_seq) merge_seq = h.seq;
@@ -533,6 +557,20 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   }
   uint64_t T = sh_term_seq[0], M = sh_merge_seq[0], RD = sh_rd_seq[0];
   uint64_t ref = sh_term_ref[0];
+  uint8_t term_type = 0xFF;
+  wb::RecHdr h = {};
+  RunView rv = {};
+  if (ref != ~0ULL) {
+    rv = runs[ref >> 32];
+    h = ((const wb::RecHdr *)(store + rv.hdr_off))[(uint32_t)ref];
+    term_type = h.type;
+  }
+  if (tid == 0 && extra) { /* raw verdict for host-side merging */
+    extra[q].term_seq = T;
+    extra[q].merge_seq = M;
+    extra[q].rd_seq = RD;
+    extra[q].term_type = term_type;
+  }
   uint64_t floor_seq = T > RD ? T : RD;
   if (M > floor_seq) { /* live merge operands: fold on the host */
     if (tid == 0) {
@@ -548,9 +586,6 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
     }
     return;
   }
-  RunView rv = runs[ref >> 32];
-  const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
-  wb::RecHdr h = hdrs[(uint32_t)ref];
   if (h.type != wb::kValue) { /* Delete/SingleDelete terminator */
     if (tid == 0) {
       out[q].status = GRA_GET_MISS;
@@ -742,7 +777,9 @@ struct GraEngine {
     GraKeyRef *d_keys = nullptr;
     uint8_t *d_keybuf = nullptr, *d_valbuf = nullptr;
     GraGetResult *d_out = nullptr;
-    size_t runs_cap = 0, keys_cap = 0, keybuf_cap = 0, valbuf_cap = 0, out_cap = 0;
+    void *d_extra = nullptr; /* MgExtra[] for mixed-shard merging */
+    size_t runs_cap = 0, keys_cap = 0, keybuf_cap = 0, valbuf_cap = 0,
+           out_cap = 0, extra_cap = 0;
   } mg;
   /* drain-host pinned arena pool. DECLARED BEFORE shards: runs hold
    * shared_ptrs whose deleter returns buffers here, so the pool must be
@@ -979,7 +1016,7 @@ GraEngine::~GraEngine() {
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
   for (void *p : {(void *)mg.d_runs, (void *)mg.d_keys, (void *)mg.d_keybuf,
-                  (void *)mg.d_valbuf, (void *)mg.d_out})
+                  (void *)mg.d_valbuf, (void *)mg.d_out, mg.d_extra})
     if (p) (void)hipFree(p);
   for (int i = 0; i < 2; i++) {
     if (d_stage_blobs_bufs[i]) (void)hipFree(d_stage_blobs_bufs[i]);
@@ -1904,33 +1941,34 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
   GraEngine *e = db->e;
   if (nq == 0) return GRA_OK;
   ShardState &ss = e->shards[db->shard];
-  /* snapshot the device-resident run list (newest->oldest for the kernel);
-   * host-origin runs (leader writes) have no device bytes — fall back to the
-   * host path for the whole call when any exist (mixed shards are the
-   * exception, not the serving case) */
+  /* snapshot the run list: device-resident runs feed the kernel; host-only
+   * runs (leader writes, ring+drain arenas) are probed on the host and the
+   * two halves merge by seq — a mixed shard no longer forces the whole
+   * call to the host path */
   std::vector<RunView> views;
+  std::vector<std::shared_ptr<Run>> host_runs;
   {
     std::lock_guard<std::mutex> lk(ss.mu);
     views.reserve(ss.runs.size());
     for (auto it = ss.runs.rbegin(); it != ss.runs.rend(); ++it) {
       const Run &r = **it;
       if (r.n_entries == 0) continue;
-      if (r.hdr_cur == UINT64_MAX) { /* host-origin run */
-        views.clear();
-        break;
+      if (r.hdr_cur == UINT64_MAX) {
+        host_runs.push_back(*it); /* oldest..newest order irrelevant: probe
+                                     tracks max seqs */
+      } else {
+        views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
       }
-      views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
-    }
-    if (views.empty() && !ss.runs.empty()) {
-      /* host fallback: answer every query via gra_get */
-      for (uint32_t q = 0; q < nq; q++) out[q].status = GRA_GET_NEEDS_HOST;
-      return GRA_OK;
     }
   }
   if (views.empty()) {
-    for (uint32_t q = 0; q < nq; q++) {
-      out[q].status = GRA_GET_MISS;
-      out[q].vlen = 0;
+    if (host_runs.empty()) {
+      for (uint32_t q = 0; q < nq; q++) {
+        out[q].status = GRA_GET_MISS;
+        out[q].vlen = 0;
+      }
+    } else { /* purely host-resident shard: answer via the host path */
+      for (uint32_t q = 0; q < nq; q++) out[q].status = GRA_GET_NEEDS_HOST;
     }
     return GRA_OK;
   }
@@ -1961,6 +1999,12 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
   GraKeyRef *d_keys = mg.d_keys;
   uint8_t *d_keybuf = mg.d_keybuf, *d_valbuf = mg.d_valbuf;
   GraGetResult *d_out = mg.d_out;
+  const bool mixed = !host_runs.empty();
+  std::vector<MgExtra> h_extra;
+  if (mixed && !grow(&mg.d_extra, &mg.extra_cap, nq * sizeof(MgExtra))) {
+    g_err = "gra_multiget: allocation failed";
+    return GRA_ERR;
+  }
   int rc = GRA_ERR;
   do {
     if (hipMemcpy(d_runs, views.data(), views.size() * sizeof(RunView),
@@ -1972,16 +2016,66 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
       break;
     hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream, e->d_store,
                        d_runs, (uint32_t)views.size(), d_keys, d_keybuf, nq,
-                       d_valbuf, val_stride, d_out);
+                       d_valbuf, val_stride, d_out,
+                       mixed ? (MgExtra *)mg.d_extra : nullptr);
     if (hipGetLastError() != hipSuccess) break;
     if (hipStreamSynchronize(e->stream) != hipSuccess) break;
     if (hipMemcpy(out, d_out, nq * sizeof(GraGetResult),
                   hipMemcpyDeviceToHost) != hipSuccess ||
         hipMemcpy(valbuf, d_valbuf, vb, hipMemcpyDeviceToHost) != hipSuccess)
       break;
+    if (mixed) {
+      h_extra.resize(nq);
+      if (hipMemcpy(h_extra.data(), mg.d_extra, nq * sizeof(MgExtra),
+                    hipMemcpyDeviceToHost) != hipSuccess)
+        break;
+    }
     rc = GRA_OK;
   } while (0);
-  if (rc != GRA_OK) g_err = "gra_multiget: device op failed";
+  if (rc != GRA_OK) {
+    g_err = "gra_multiget: device op failed";
+    return rc;
+  }
+  if (mixed) {
+    /* merge the device verdict with a host-run probe, by seq — exactly the
+     * k_multiget decision re-run over the union */
+    for (uint32_t q = 0; q < nq; q++) {
+      if (out[q].status == GRA_GET_NEEDS_HOST) continue; /* full fold */
+      ProbeResult hp;
+      run_probe(host_runs, keybuf + keys[q].off, keys[q].len, &hp);
+      const MgExtra &de = h_extra[q];
+      uint64_t T = de.term_seq, M = de.merge_seq, RD = de.rd_seq;
+      bool host_wins = hp.term_seq > T;
+      if (host_wins) T = hp.term_seq;
+      if (hp.merge_seq > M) M = hp.merge_seq;
+      if (hp.rd_seq > RD) RD = hp.rd_seq;
+      uint64_t fl = T > RD ? T : RD;
+      if (M > fl) {
+        out[q].status = GRA_GET_NEEDS_HOST; /* cross-half fold */
+        out[q].vlen = 0;
+        continue;
+      }
+      if (T == 0 || T <= RD) {
+        out[q].status = GRA_GET_MISS;
+        out[q].vlen = 0;
+        continue;
+      }
+      if (host_wins) {
+        if (hp.term_type != wb::kValue) {
+          out[q].status = GRA_GET_MISS;
+          out[q].vlen = 0;
+        } else {
+          uint32_t v = hp.vlen < val_stride ? hp.vlen : val_stride;
+          memcpy(valbuf + (size_t)q * val_stride, hp.val, v);
+          out[q].status = GRA_GET_FOUND;
+          out[q].vlen = hp.vlen;
+        }
+      } else if (de.term_type != wb::kValue || T == 0) {
+        out[q].status = GRA_GET_MISS;
+        out[q].vlen = 0;
+      } /* else: the device verdict (+ value already in valbuf) stands */
+    }
+  }
   return rc;
 }
 

@@ -86,6 +86,38 @@ int run_get(const std::vector<std::shared_ptr<Run>> &runs, const void *key_,
   return 0;
 }
 
+void run_probe(const std::vector<std::shared_ptr<Run>> &runs, const void *key_,
+               size_t klen, ProbeResult *out) {
+  const uint8_t *key = (const uint8_t *)key_;
+  *out = ProbeResult();
+  for (const auto &rp : runs) {
+    const Run &r = *rp;
+    if (r.n_entries == 0) continue;
+    const wb::RecHdr *hdrs = (const wb::RecHdr *)r.hdrs_data();
+    const uint8_t *pay = r.payload_data();
+    for (uint32_t i = 0; i < r.n_entries; i++) {
+      const wb::RecHdr &h = hdrs[i];
+      if (h.type == wb::kRangeDeletion) {
+        if (range_covers(pay + h.kv_off, h.key_len,
+                         pay + h.kv_off + h.key_len, h.val_len, key, klen) &&
+            h.seq > out->rd_seq)
+          out->rd_seq = h.seq;
+        continue;
+      }
+      if (h.key_len != klen || memcmp(pay + h.kv_off, key, klen) != 0)
+        continue;
+      if (h.type == wb::kMerge) {
+        if (h.seq > out->merge_seq) out->merge_seq = h.seq;
+      } else if (h.seq > out->term_seq) {
+        out->term_seq = h.seq;
+        out->term_type = h.type;
+        out->val = pay + h.kv_off + h.key_len;
+        out->vlen = h.val_len;
+      }
+    }
+  }
+}
+
 bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out) {
   wb::WalkTotals tot = wb::walk(rep, (uint32_t)len, nullptr, 0);
   if (!tot.ok) return false;
@@ -125,7 +157,8 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
-      h.kpref = 0; /* reserved (see k_emit) */
+      /* same stored-key prefix the GPU path fills (cf id for cf records) */
+      h.kpref = cf4 ? rc.cf_id : wb::key_prefix4(rep + rc.key_off, rc.key_len);
       hd[i] = h;
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);

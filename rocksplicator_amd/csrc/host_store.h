@@ -86,6 +86,19 @@ struct ShardState {
 int run_get(const std::vector<std::shared_ptr<Run>> &runs, const void *key,
             size_t klen, int merge_op, std::string *out);
 
+/* Raw probe over a run list: max-seq terminator (Put/Delete/SingleDelete),
+ * max merge-operand seq and max covering range-tombstone seq for `key` —
+ * the same per-query verdict k_multiget computes, so a mixed shard can
+ * merge the device and host halves by seq. */
+struct ProbeResult {
+  uint64_t term_seq = 0, merge_seq = 0, rd_seq = 0;
+  uint8_t term_type = 0xFF;          /* wb tag of the terminator */
+  const uint8_t *val = nullptr;      /* terminator value (kValue only) */
+  uint32_t vlen = 0;
+};
+void run_probe(const std::vector<std::shared_ptr<Run>> &runs, const void *key,
+               size_t klen, ProbeResult *out);
+
 /* Host apply of one rep blob (leader write path): decodes with wb::walk and
  * builds a Run in the same format the GPU emits. Returns false on corrupt
  * rep. base_seq = first seq to assign. */

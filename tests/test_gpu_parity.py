@@ -1007,3 +1007,39 @@ def test_drain_streaming_with_corruption(olib):
     assert db.get(b"x") is None
     assert db.checksum() == olib_checksum(olib, ost)
     e.close()
+
+
+def test_multiget_mixed_host_device_runs(olib):
+    """Leader-written host runs and GPU-applied device runs interleave in one
+    shard; gra_multiget now merges both halves by seq instead of falling
+    back wholesale. Every outcome must equal gra_get (which equals the
+    oracle)."""
+    e = ra.Engine(nshards=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1)
+
+    def both(rep, via):
+        if via == "host":
+            db.write_leader(rep)
+        else:
+            assert db.handle_replicate_response(rep)
+            e.flush()
+        assert ost.apply(0, rep)
+
+    both(PyBatch().put(b"a", b"dev1").put(b"b", b"dev1").data(), "gpu")
+    both(PyBatch().put(b"a", b"host2").delete(b"b").data(), "host")
+    both(PyBatch().put(b"c", b"dev3").put(b"b", b"dev3").data(), "gpu")
+    both(PyBatch().delete(b"c").merge(b"m", b"x").data(), "host")
+    both(PyBatch().merge(b"m", b"y").delete_range(b"a", b"ab").data(), "gpu")
+    probes = [b"a", b"b", b"c", b"m", b"zz"]
+    assert db.latest_seq() == ost.latest_seq(0)
+    for k in probes:
+        assert db.get(k) == ost.get(0, k), k
+    for k, v in zip(probes, db.multiget(probes)):
+        assert v == ost.get(0, k), k
+    # newer device run over an older host terminator and vice versa
+    both(PyBatch().put(b"b", b"dev-final").data(), "gpu")
+    both(PyBatch().put(b"a", b"host-final").data(), "host")
+    for k, v in zip([b"a", b"b"], db.multiget([b"a", b"b"])):
+        assert v == ost.get(0, k), k
+    e.close()
