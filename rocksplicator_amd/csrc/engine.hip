@@ -283,11 +283,9 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = btag;
     h.flags = cf4 ? 1 : 0;
-    /* kpref = first 4 STORED key bytes (cf records: the cf id itself —
-     * free here; plain records: k_copy fills it from the bytes it is
-     * copying anyway). Lets k_multiget reject non-matching entries on the
-     * header alone, no payload touch. */
-    h.kpref = cf4 ? r.cf_id : 0;
+    h.kpref = 0; /* k_copy fills the key fingerprint from the bytes it is
+                    streaming anyway; lets k_multiget reject non-matching
+                    entries on the header alone, no payload touch */
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -395,11 +393,15 @@ __global__ void __launch_bounds__(256) k_copy(const uint8_t *__restrict__ blobs,
   for (uint32_t t = g; t < ntasks; t += ngroups) {
     CopyTask tk = tasks[t];
     uint32_t nb = tk.nbytes & 0x7FFFFFFFu;
-    if ((t & 1) == 0 && lane == 0 && !(tk.nbytes >> 31)) {
-      /* plain-key task: fill the header's key prefix from the bytes this
-       * group is about to stream anyway (same cachelines — near free; the
-       * r01 emit-side attempt re-touched COLD blob bytes and lost) */
-      hdrs[t >> 1].kpref = wb::key_prefix4(blobs + tk.src_off, nb);
+    if ((t & 1) == 0 && lane == 0) {
+      /* key task: fingerprint the STORED key from bytes this group is
+       * about to stream anyway (same cachelines — near free; the r01
+       * emit-side attempt re-touched COLD blob bytes and lost). For cf
+       * records fold the 4 prefix bytes k_emit just wrote first. */
+      uint32_t fh = wb::kFnvBasis32;
+      if (tk.nbytes >> 31)
+        fh = wb::key_fnv_fold(fh, pay_region + tk.dst_rel - 4, 4);
+      hdrs[t >> 1].kpref = wb::key_fnv_fold(fh, blobs + tk.src_off, nb);
     }
     if (nb == 0) continue;
     copy_dwordx4<G>(pay_region + tk.dst_rel, blobs + tk.src_off, nb, lane);
@@ -504,7 +506,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   if (q >= nq) return;
   const uint8_t *key = keybuf + keys[q].off;
   uint32_t klen = keys[q].len;
-  uint32_t qpref = wb::key_prefix4(key, klen);
+  uint32_t qpref = wb::key_fnv_fold(wb::kFnvBasis32, key, klen);
   uint64_t term_seq = 0, term_ref = ~0ULL, merge_seq = 0, rd_seq = 0;
   for (uint32_t r = 0; r < nruns; r++) {
     RunView rv = runs[r];

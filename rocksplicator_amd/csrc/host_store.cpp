@@ -157,8 +157,12 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
-      /* same stored-key prefix the GPU path fills (cf id for cf records) */
-      h.kpref = cf4 ? rc.cf_id : wb::key_prefix4(rep + rc.key_off, rc.key_len);
+      /* same stored-key fingerprint the GPU copy kernel fills */
+      {
+        uint32_t fh = wb::kFnvBasis32;
+        if (cf4) fh = wb::key_fnv_fold(fh, (const uint8_t *)&rc.cf_id, 4);
+        h.kpref = wb::key_fnv_fold(fh, rep + rc.key_off, rc.key_len);
+      }
       hd[i] = h;
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);
