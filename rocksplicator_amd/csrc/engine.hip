@@ -606,6 +606,167 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   }
 }
 
+/* ---------------- hash-join multiget ----------------
+ * The linear per-query scan is header-bandwidth bound (nq × entries × 24 B).
+ * Invert it: ONE coalesced pass over all entries probes an open-addressed
+ * query-fingerprint table (kpref -> query idx) and appends the rare
+ * matches to a candidate list; tiny resolve kernels then pick per-query
+ * winners by seq. Traffic: O(entries + matches) instead of O(nq × entries).
+ * Range tombstones are collected in the same pass and tested per query
+ * separately (they cover key RANGES — not hash-matchable). Overflow of
+ * either list falls back to the per-query scan kernel. */
+struct MgCand {
+  uint32_t qidx, run, entry, type;
+  uint64_t seq;
+};
+struct MgTomb {
+  uint32_t run, entry, _pad0, _pad1;
+  uint64_t seq;
+};
+
+__global__ void k_mg_scan(const uint8_t *__restrict__ store,
+                          const RunView *__restrict__ runs,
+                          const uint64_t *__restrict__ qtab, uint32_t qmask,
+                          const GraKeyRef *__restrict__ keys,
+                          const uint8_t *__restrict__ keybuf,
+                          MgCand *__restrict__ cands,
+                          uint32_t *__restrict__ ncand, uint32_t cand_cap,
+                          MgTomb *__restrict__ tombs,
+                          uint32_t *__restrict__ ntomb, uint32_t tomb_cap) {
+  uint32_t r = blockIdx.y;
+  RunView rv = runs[r];
+  const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+  const uint8_t *pay = store + rv.payload_off;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < rv.n_entries;
+       i += gridDim.x * blockDim.x) {
+    wb::RecHdr h = hdrs[i];
+    if (h.type == wb::kRangeDeletion) {
+      uint32_t slot = atomicAdd(ntomb, 1u);
+      if (slot < tomb_cap) tombs[slot] = {r, i, 0, 0, h.seq};
+      continue;
+    }
+    uint32_t s = h.kpref & qmask;
+    while (true) { /* linear probe; duplicates of a fingerprint chain on */
+      uint64_t e = qtab[s];
+      if (e == 0) break;
+      if ((uint32_t)(e >> 32) == h.kpref) {
+        uint32_t qidx = (uint32_t)e - 1;
+        uint32_t klen = keys[qidx].len;
+        if (h.key_len == klen &&
+            dev_memcmp(pay + (h.kv_off - rv.pay_rel_base),
+                       keybuf + keys[qidx].off, klen) == 0) {
+          uint32_t slot = atomicAdd(ncand, 1u);
+          if (slot < cand_cap) cands[slot] = {qidx, r, i, h.type, h.seq};
+        }
+      }
+      s = (s + 1) & qmask;
+    }
+  }
+}
+
+__global__ void k_mg_resolve1(const MgCand *__restrict__ cands, uint32_t n,
+                              unsigned long long *__restrict__ term_pack,
+                              unsigned long long *__restrict__ merge_seq) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  MgCand c = cands[i];
+  if (c.type == wb::kMerge)
+    atomicMax(&merge_seq[c.qidx], (unsigned long long)c.seq);
+  else /* seqs are unique per shard: the packed max has ONE winner */
+    atomicMax(&term_pack[c.qidx],
+              (unsigned long long)((c.seq << 8) | c.type));
+}
+
+__global__ void k_mg_tombs(const uint8_t *__restrict__ store,
+                           const RunView *__restrict__ runs,
+                           const MgTomb *__restrict__ tombs, uint32_t ntomb,
+                           const GraKeyRef *__restrict__ keys,
+                           const uint8_t *__restrict__ keybuf, uint32_t nq,
+                           unsigned long long *__restrict__ rd_seq) {
+  uint32_t q = blockIdx.x * blockDim.x + threadIdx.x;
+  if (q >= nq) return;
+  const uint8_t *key = keybuf + keys[q].off;
+  uint32_t klen = keys[q].len;
+  uint64_t best = 0;
+  for (uint32_t t = 0; t < ntomb; t++) {
+    MgTomb mt = tombs[t];
+    RunView rv = runs[mt.run];
+    const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+    wb::RecHdr h = hdrs[mt.entry];
+    const uint8_t *pay = store + rv.payload_off;
+    uint32_t rel = h.kv_off - rv.pay_rel_base;
+    const uint8_t *b = pay + rel, *e2 = pay + rel + h.key_len;
+    uint32_t bl = h.key_len, el = h.val_len;
+    int c1 = dev_memcmp(b, key, bl < klen ? bl : klen);
+    if (c1 > 0 || (c1 == 0 && bl > klen)) continue;
+    int c2 = dev_memcmp(key, e2, klen < el ? klen : el);
+    if (c2 > 0 || (c2 == 0 && klen >= el)) continue;
+    if (h.seq > best) best = h.seq;
+  }
+  rd_seq[q] = best;
+}
+
+__global__ void k_mg_resolve2(const MgCand *__restrict__ cands, uint32_t n,
+                              const unsigned long long *__restrict__ term_pack,
+                              unsigned long long *__restrict__ winner_ref) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  MgCand c = cands[i];
+  if (c.type != wb::kMerge &&
+      term_pack[c.qidx] == (unsigned long long)((c.seq << 8) | c.type))
+    winner_ref[c.qidx] = ((unsigned long long)c.run << 32) | c.entry;
+}
+
+__global__ void k_mg_emit(const uint8_t *__restrict__ store,
+                          const RunView *__restrict__ runs,
+                          const unsigned long long *__restrict__ term_pack,
+                          const unsigned long long *__restrict__ merge_seq,
+                          const unsigned long long *__restrict__ rd_seq,
+                          const unsigned long long *__restrict__ winner_ref,
+                          uint32_t nq, uint8_t *__restrict__ valbuf,
+                          uint32_t val_stride, GraGetResult *__restrict__ out,
+                          MgExtra *__restrict__ extra) {
+  uint32_t q = blockIdx.x;
+  if (q >= nq) return;
+  uint64_t tp = term_pack[q];
+  uint64_t T = tp >> 8, M = merge_seq[q], RD = rd_seq[q];
+  uint8_t ttype = tp ? (uint8_t)tp : 0xFF;
+  if (threadIdx.x == 0 && extra) {
+    extra[q].term_seq = T;
+    extra[q].merge_seq = M;
+    extra[q].rd_seq = RD;
+    extra[q].term_type = ttype;
+  }
+  uint64_t fl = T > RD ? T : RD;
+  if (M > fl) {
+    if (threadIdx.x == 0) {
+      out[q].status = GRA_GET_NEEDS_HOST;
+      out[q].vlen = 0;
+    }
+    return;
+  }
+  if (T == 0 || T <= RD || ttype != wb::kValue) {
+    if (threadIdx.x == 0) {
+      out[q].status = GRA_GET_MISS;
+      out[q].vlen = 0;
+    }
+    return;
+  }
+  unsigned long long ref = winner_ref[q];
+  RunView rv = runs[ref >> 32];
+  const wb::RecHdr *hdrs = (const wb::RecHdr *)(store + rv.hdr_off);
+  wb::RecHdr h = hdrs[(uint32_t)ref];
+  uint32_t vlen = h.val_len < val_stride ? h.val_len : val_stride;
+  const uint8_t *src =
+      store + rv.payload_off + (h.kv_off - rv.pay_rel_base) + h.key_len;
+  uint8_t *dst = valbuf + (size_t)q * val_stride;
+  for (uint32_t b = threadIdx.x; b < vlen; b += blockDim.x) dst[b] = src[b];
+  if (threadIdx.x == 0) {
+    out[q].status = GRA_GET_FOUND;
+    out[q].vlen = h.val_len;
+  }
+}
+
 /* ---- full-store checksum (parity at any size) ----
  * Same record hash as the oracle's orc_shard_checksum: FNV-1a over
  * (seq LE8 | type | key_len LE4 | val_len LE4 | key | val), u64-ADD
@@ -780,8 +941,15 @@ struct GraEngine {
     uint8_t *d_keybuf = nullptr, *d_valbuf = nullptr;
     GraGetResult *d_out = nullptr;
     void *d_extra = nullptr; /* MgExtra[] for mixed-shard merging */
+    void *d_qtab = nullptr;  /* hash-join: query fingerprint table (u64) */
+    void *d_cands = nullptr; /* MgCand[] matches */
+    void *d_tombs = nullptr; /* MgTomb[] range tombstones seen in the scan */
+    void *d_aux = nullptr;   /* 4 x nq u64: term_pack/merge/rd/winner */
+    uint32_t *d_counts = nullptr; /* {ncand, ntomb} */
+    uint32_t *h_counts = nullptr; /* pinned mirror */
     size_t runs_cap = 0, keys_cap = 0, keybuf_cap = 0, valbuf_cap = 0,
-           out_cap = 0, extra_cap = 0;
+           out_cap = 0, extra_cap = 0, qtab_cap = 0, cands_cap = 0,
+           tombs_cap = 0, aux_cap = 0;
   } mg;
   /* drain-host pinned arena pool. DECLARED BEFORE shards: runs hold
    * shared_ptrs whose deleter returns buffers here, so the pool must be
@@ -1018,8 +1186,11 @@ GraEngine::~GraEngine() {
     if (s.h_descs) (void)hipHostFree(s.h_descs);
   }
   for (void *p : {(void *)mg.d_runs, (void *)mg.d_keys, (void *)mg.d_keybuf,
-                  (void *)mg.d_valbuf, (void *)mg.d_out, mg.d_extra})
+                  (void *)mg.d_valbuf, (void *)mg.d_out, mg.d_extra,
+                  mg.d_qtab, mg.d_cands, mg.d_tombs, mg.d_aux,
+                  (void *)mg.d_counts})
     if (p) (void)hipFree(p);
+  if (mg.h_counts) (void)hipHostFree(mg.h_counts);
   for (int i = 0; i < 2; i++) {
     if (d_stage_blobs_bufs[i]) (void)hipFree(d_stage_blobs_bufs[i]);
     if (d_stage_descs_bufs[i]) (void)hipFree(d_stage_descs_bufs[i]);
@@ -2003,9 +2174,38 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
   GraGetResult *d_out = mg.d_out;
   const bool mixed = !host_runs.empty();
   std::vector<MgExtra> h_extra;
-  if (mixed && !grow(&mg.d_extra, &mg.extra_cap, nq * sizeof(MgExtra))) {
+  /* hash-join path setup: query fingerprint table + candidate buffers */
+  uint32_t qtab_size = 64;
+  while (qtab_size < 2 * nq) qtab_size <<= 1;
+  uint32_t cand_cap = 4 * nq + 1024, tomb_cap = 4096;
+  bool hashjoin = views.size() <= 65535;
+  if ((mixed && !grow(&mg.d_extra, &mg.extra_cap, nq * sizeof(MgExtra))) ||
+      (hashjoin &&
+       (!grow(&mg.d_qtab, &mg.qtab_cap, qtab_size * 8) ||
+        !grow(&mg.d_cands, &mg.cands_cap, (size_t)cand_cap * sizeof(MgCand)) ||
+        !grow(&mg.d_tombs, &mg.tombs_cap, (size_t)tomb_cap * sizeof(MgTomb)) ||
+        !grow(&mg.d_aux, &mg.aux_cap, (size_t)nq * 8 * 4)))) {
     g_err = "gra_multiget: allocation failed";
     return GRA_ERR;
+  }
+  if (!mg.d_counts) {
+    if (hipMalloc(&mg.d_counts, 8) != hipSuccess ||
+        hipHostMalloc(&mg.h_counts, 8) != hipSuccess) {
+      g_err = "gra_multiget: allocation failed";
+      return GRA_ERR;
+    }
+  }
+  std::vector<uint64_t> qtab;
+  if (hashjoin) { /* host-built open-addressed fingerprint table */
+    qtab.assign(qtab_size, 0);
+    uint32_t mask = qtab_size - 1;
+    for (uint32_t q = 0; q < nq; q++) {
+      uint32_t fp =
+          wb::key_fnv_fold(wb::kFnvBasis32, keybuf + keys[q].off, keys[q].len);
+      uint32_t s = fp & mask;
+      while (qtab[s] != 0) s = (s + 1) & mask;
+      qtab[s] = ((uint64_t)fp << 32) | (q + 1);
+    }
   }
   int rc = GRA_ERR;
   do {
@@ -2016,11 +2216,64 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
         hipMemcpy(d_keybuf, keybuf, keybuf_len, hipMemcpyHostToDevice) !=
             hipSuccess)
       break;
-    hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream, e->d_store,
-                       d_runs, (uint32_t)views.size(), d_keys, d_keybuf, nq,
-                       d_valbuf, val_stride, d_out,
-                       mixed ? (MgExtra *)mg.d_extra : nullptr);
-    if (hipGetLastError() != hipSuccess) break;
+    bool done = false;
+    if (hashjoin) {
+      uint32_t nruns32 = (uint32_t)views.size();
+      if (hipMemcpy(mg.d_qtab, qtab.data(), qtab_size * 8,
+                    hipMemcpyHostToDevice) != hipSuccess ||
+          hipMemsetAsync(mg.d_counts, 0, 8, e->stream) != hipSuccess ||
+          hipMemsetAsync(mg.d_aux, 0, (size_t)nq * 8 * 4, e->stream) !=
+              hipSuccess)
+        break;
+      uint64_t max_entries = 0;
+      for (auto &v : views)
+        max_entries = v.n_entries > max_entries ? v.n_entries : max_entries;
+      uint32_t bx = (uint32_t)((max_entries + 255) / 256);
+      if (bx > 1024) bx = 1024;
+      if (bx == 0) bx = 1;
+      hipLaunchKernelGGL(k_mg_scan, dim3(bx, nruns32), dim3(256), 0, e->stream,
+                         e->d_store, d_runs, (const uint64_t *)mg.d_qtab,
+                         qtab_size - 1, d_keys, d_keybuf, (MgCand *)mg.d_cands,
+                         mg.d_counts, cand_cap, (MgTomb *)mg.d_tombs,
+                         mg.d_counts + 1, tomb_cap);
+      if (hipGetLastError() != hipSuccess) break;
+      if (hipMemcpyAsync(mg.h_counts, mg.d_counts, 8, hipMemcpyDeviceToHost,
+                         e->stream) != hipSuccess ||
+          hipStreamSynchronize(e->stream) != hipSuccess)
+        break;
+      uint32_t ncand = mg.h_counts[0], ntomb = mg.h_counts[1];
+      if (ncand <= cand_cap && ntomb <= tomb_cap) {
+        unsigned long long *aux = (unsigned long long *)mg.d_aux;
+        unsigned long long *term_pack = aux, *mergeq = aux + nq,
+                           *rdq = aux + 2 * nq, *winner = aux + 3 * nq;
+        if (ncand)
+          hipLaunchKernelGGL(k_mg_resolve1, dim3((ncand + 255) / 256),
+                             dim3(256), 0, e->stream, (MgCand *)mg.d_cands,
+                             ncand, term_pack, mergeq);
+        if (ntomb)
+          hipLaunchKernelGGL(k_mg_tombs, dim3((nq + 255) / 256), dim3(256), 0,
+                             e->stream, e->d_store, d_runs,
+                             (MgTomb *)mg.d_tombs, ntomb, d_keys, d_keybuf,
+                             nq, rdq);
+        if (ncand)
+          hipLaunchKernelGGL(k_mg_resolve2, dim3((ncand + 255) / 256),
+                             dim3(256), 0, e->stream, (MgCand *)mg.d_cands,
+                             ncand, term_pack, winner);
+        hipLaunchKernelGGL(k_mg_emit, dim3(nq), dim3(64), 0, e->stream,
+                           e->d_store, d_runs, term_pack, mergeq, rdq, winner,
+                           nq, d_valbuf, val_stride, d_out,
+                           mixed ? (MgExtra *)mg.d_extra : nullptr);
+        if (hipGetLastError() != hipSuccess) break;
+        done = true;
+      } /* overflow: fall through to the per-query scan */
+    }
+    if (!done) {
+      hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream,
+                         e->d_store, d_runs, (uint32_t)views.size(), d_keys,
+                         d_keybuf, nq, d_valbuf, val_stride, d_out,
+                         mixed ? (MgExtra *)mg.d_extra : nullptr);
+      if (hipGetLastError() != hipSuccess) break;
+    }
     if (hipStreamSynchronize(e->stream) != hipSuccess) break;
     if (hipMemcpy(out, d_out, nq * sizeof(GraGetResult),
                   hipMemcpyDeviceToHost) != hipSuccess ||
