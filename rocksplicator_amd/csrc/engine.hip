@@ -664,9 +664,14 @@ __global__ void k_mg_scan(const uint8_t *__restrict__ store,
   }
 }
 
-__global__ void k_mg_resolve1(const MgCand *__restrict__ cands, uint32_t n,
+__global__ void k_mg_resolve1(const MgCand *__restrict__ cands,
+                              const uint32_t *__restrict__ ncand_p,
+                              uint32_t cap,
                               unsigned long long *__restrict__ term_pack,
                               unsigned long long *__restrict__ merge_seq) {
+  uint32_t n = *ncand_p < cap ? *ncand_p : cap; /* device-side bound: no
+                                                   host sync between scan
+                                                   and resolve */
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   MgCand c = cands[i];
@@ -679,10 +684,12 @@ __global__ void k_mg_resolve1(const MgCand *__restrict__ cands, uint32_t n,
 
 __global__ void k_mg_tombs(const uint8_t *__restrict__ store,
                            const RunView *__restrict__ runs,
-                           const MgTomb *__restrict__ tombs, uint32_t ntomb,
+                           const MgTomb *__restrict__ tombs,
+                           const uint32_t *__restrict__ ntomb_p, uint32_t cap,
                            const GraKeyRef *__restrict__ keys,
                            const uint8_t *__restrict__ keybuf, uint32_t nq,
                            unsigned long long *__restrict__ rd_seq) {
+  uint32_t ntomb = *ntomb_p < cap ? *ntomb_p : cap;
   uint32_t q = blockIdx.x * blockDim.x + threadIdx.x;
   if (q >= nq) return;
   const uint8_t *key = keybuf + keys[q].off;
@@ -706,9 +713,12 @@ __global__ void k_mg_tombs(const uint8_t *__restrict__ store,
   rd_seq[q] = best;
 }
 
-__global__ void k_mg_resolve2(const MgCand *__restrict__ cands, uint32_t n,
+__global__ void k_mg_resolve2(const MgCand *__restrict__ cands,
+                              const uint32_t *__restrict__ ncand_p,
+                              uint32_t cap,
                               const unsigned long long *__restrict__ term_pack,
                               unsigned long long *__restrict__ winner_ref) {
+  uint32_t n = *ncand_p < cap ? *ncand_p : cap;
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   MgCand c = cands[i];
@@ -2208,85 +2218,89 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
     }
   }
   int rc = GRA_ERR;
-  do {
-    if (hipMemcpy(d_runs, views.data(), views.size() * sizeof(RunView),
-                  hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemcpy(d_keys, keys, nq * sizeof(GraKeyRef),
-                  hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemcpy(d_keybuf, keybuf, keybuf_len, hipMemcpyHostToDevice) !=
-            hipSuccess)
-      break;
-    bool done = false;
-    if (hashjoin) {
-      uint32_t nruns32 = (uint32_t)views.size();
-      if (hipMemcpy(mg.d_qtab, qtab.data(), qtab_size * 8,
-                    hipMemcpyHostToDevice) != hipSuccess ||
-          hipMemsetAsync(mg.d_counts, 0, 8, e->stream) != hipSuccess ||
-          hipMemsetAsync(mg.d_aux, 0, (size_t)nq * 8 * 4, e->stream) !=
-              hipSuccess)
+  for (int attempt = 0; attempt < 2 && rc != GRA_OK; attempt++) {
+    /* attempt 0: hash-join, fully async, ONE sync at the end. attempt 1
+     * (only when the candidate/tombstone caps overflowed): the per-query
+     * scan kernel. */
+    bool use_hj = hashjoin && attempt == 0;
+    do {
+      if (hipMemcpyAsync(d_runs, views.data(),
+                         views.size() * sizeof(RunView),
+                         hipMemcpyHostToDevice, e->stream) != hipSuccess ||
+          hipMemcpyAsync(d_keys, keys, nq * sizeof(GraKeyRef),
+                         hipMemcpyHostToDevice, e->stream) != hipSuccess ||
+          hipMemcpyAsync(d_keybuf, keybuf, keybuf_len,
+                         hipMemcpyHostToDevice, e->stream) != hipSuccess)
         break;
-      uint64_t max_entries = 0;
-      for (auto &v : views)
-        max_entries = v.n_entries > max_entries ? v.n_entries : max_entries;
-      uint32_t bx = (uint32_t)((max_entries + 255) / 256);
-      if (bx > 1024) bx = 1024;
-      if (bx == 0) bx = 1;
-      hipLaunchKernelGGL(k_mg_scan, dim3(bx, nruns32), dim3(256), 0, e->stream,
-                         e->d_store, d_runs, (const uint64_t *)mg.d_qtab,
-                         qtab_size - 1, d_keys, d_keybuf, (MgCand *)mg.d_cands,
-                         mg.d_counts, cand_cap, (MgTomb *)mg.d_tombs,
-                         mg.d_counts + 1, tomb_cap);
-      if (hipGetLastError() != hipSuccess) break;
-      if (hipMemcpyAsync(mg.h_counts, mg.d_counts, 8, hipMemcpyDeviceToHost,
-                         e->stream) != hipSuccess ||
-          hipStreamSynchronize(e->stream) != hipSuccess)
-        break;
-      uint32_t ncand = mg.h_counts[0], ntomb = mg.h_counts[1];
-      if (ncand <= cand_cap && ntomb <= tomb_cap) {
+      if (use_hj) {
+        uint32_t nruns32 = (uint32_t)views.size();
+        if (hipMemcpyAsync(mg.d_qtab, qtab.data(), qtab_size * 8,
+                           hipMemcpyHostToDevice, e->stream) != hipSuccess ||
+            hipMemsetAsync(mg.d_counts, 0, 8, e->stream) != hipSuccess ||
+            hipMemsetAsync(mg.d_aux, 0, (size_t)nq * 8 * 4, e->stream) !=
+                hipSuccess)
+          break;
+        uint64_t max_entries = 0;
+        for (auto &v : views)
+          max_entries = v.n_entries > max_entries ? v.n_entries : max_entries;
+        uint32_t bx = (uint32_t)((max_entries + 255) / 256);
+        if (bx > 1024) bx = 1024;
+        if (bx == 0) bx = 1;
+        hipLaunchKernelGGL(k_mg_scan, dim3(bx, nruns32), dim3(256), 0,
+                           e->stream, e->d_store, d_runs,
+                           (const uint64_t *)mg.d_qtab, qtab_size - 1, d_keys,
+                           d_keybuf, (MgCand *)mg.d_cands, mg.d_counts,
+                           cand_cap, (MgTomb *)mg.d_tombs, mg.d_counts + 1,
+                           tomb_cap);
+        if (hipGetLastError() != hipSuccess) break;
         unsigned long long *aux = (unsigned long long *)mg.d_aux;
         unsigned long long *term_pack = aux, *mergeq = aux + nq,
                            *rdq = aux + 2 * nq, *winner = aux + 3 * nq;
-        if (ncand)
-          hipLaunchKernelGGL(k_mg_resolve1, dim3((ncand + 255) / 256),
-                             dim3(256), 0, e->stream, (MgCand *)mg.d_cands,
-                             ncand, term_pack, mergeq);
-        if (ntomb)
-          hipLaunchKernelGGL(k_mg_tombs, dim3((nq + 255) / 256), dim3(256), 0,
-                             e->stream, e->d_store, d_runs,
-                             (MgTomb *)mg.d_tombs, ntomb, d_keys, d_keybuf,
-                             nq, rdq);
-        if (ncand)
-          hipLaunchKernelGGL(k_mg_resolve2, dim3((ncand + 255) / 256),
-                             dim3(256), 0, e->stream, (MgCand *)mg.d_cands,
-                             ncand, term_pack, winner);
+        uint32_t rb = (cand_cap + 255) / 256;
+        hipLaunchKernelGGL(k_mg_resolve1, dim3(rb), dim3(256), 0, e->stream,
+                           (MgCand *)mg.d_cands, mg.d_counts, cand_cap,
+                           term_pack, mergeq);
+        hipLaunchKernelGGL(k_mg_tombs, dim3((nq + 255) / 256), dim3(256), 0,
+                           e->stream, e->d_store, d_runs,
+                           (MgTomb *)mg.d_tombs, mg.d_counts + 1, tomb_cap,
+                           d_keys, d_keybuf, nq, rdq);
+        hipLaunchKernelGGL(k_mg_resolve2, dim3(rb), dim3(256), 0, e->stream,
+                           (MgCand *)mg.d_cands, mg.d_counts, cand_cap,
+                           term_pack, winner);
         hipLaunchKernelGGL(k_mg_emit, dim3(nq), dim3(64), 0, e->stream,
                            e->d_store, d_runs, term_pack, mergeq, rdq, winner,
                            nq, d_valbuf, val_stride, d_out,
                            mixed ? (MgExtra *)mg.d_extra : nullptr);
         if (hipGetLastError() != hipSuccess) break;
-        done = true;
-      } /* overflow: fall through to the per-query scan */
-    }
-    if (!done) {
-      hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream,
-                         e->d_store, d_runs, (uint32_t)views.size(), d_keys,
-                         d_keybuf, nq, d_valbuf, val_stride, d_out,
-                         mixed ? (MgExtra *)mg.d_extra : nullptr);
-      if (hipGetLastError() != hipSuccess) break;
-    }
-    if (hipStreamSynchronize(e->stream) != hipSuccess) break;
-    if (hipMemcpy(out, d_out, nq * sizeof(GraGetResult),
-                  hipMemcpyDeviceToHost) != hipSuccess ||
-        hipMemcpy(valbuf, d_valbuf, vb, hipMemcpyDeviceToHost) != hipSuccess)
-      break;
-    if (mixed) {
-      h_extra.resize(nq);
-      if (hipMemcpy(h_extra.data(), mg.d_extra, nq * sizeof(MgExtra),
-                    hipMemcpyDeviceToHost) != hipSuccess)
+        if (hipMemcpyAsync(mg.h_counts, mg.d_counts, 8,
+                           hipMemcpyDeviceToHost, e->stream) != hipSuccess)
+          break;
+      } else {
+        hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream,
+                           e->d_store, d_runs, (uint32_t)views.size(), d_keys,
+                           d_keybuf, nq, d_valbuf, val_stride, d_out,
+                           mixed ? (MgExtra *)mg.d_extra : nullptr);
+        if (hipGetLastError() != hipSuccess) break;
+      }
+      if (hipMemcpyAsync(out, d_out, nq * sizeof(GraGetResult),
+                         hipMemcpyDeviceToHost, e->stream) != hipSuccess ||
+          hipMemcpyAsync(valbuf, d_valbuf, vb, hipMemcpyDeviceToHost,
+                         e->stream) != hipSuccess)
         break;
-    }
-    rc = GRA_OK;
-  } while (0);
+      if (mixed) {
+        h_extra.resize(nq);
+        if (hipMemcpyAsync(h_extra.data(), mg.d_extra, nq * sizeof(MgExtra),
+                           hipMemcpyDeviceToHost, e->stream) != hipSuccess)
+          break;
+      }
+      if (hipStreamSynchronize(e->stream) != hipSuccess) break;
+      if (use_hj &&
+          (mg.h_counts[0] > cand_cap || mg.h_counts[1] > tomb_cap))
+        break; /* caps overflowed: results invalid, retry via scan kernel */
+      rc = GRA_OK;
+    } while (0);
+    if (rc != GRA_OK && use_hj) (void)hipStreamSynchronize(e->stream);
+  }
   if (rc != GRA_OK) {
     g_err = "gra_multiget: device op failed";
     return rc;
