@@ -49,23 +49,43 @@ r.sync()
 print(f"loaded {N_ENTRIES} entries, latest_seq={db.latest_seq()}")
 
 probe = [keys[rng.randrange(N_ENTRIES)] for _ in range(BATCH)]
-miss = [f"nope{i:08d}".encode() for i in range(BATCH // 8)]
+miss = [f"nope{i:08d}".encode() for i in range(BATCH)]
 
-# warmup + correctness
+# warmup + correctness (python helper path)
 res = db.multiget(probe[:64], val_stride=128)
 for k, v in zip(probe[:64], res):
     assert v == db.get(k), k
 
-t0 = time.perf_counter()
-for _ in range(ROUNDS):
-    db.multiget(probe, val_stride=128)
-dt = time.perf_counter() - t0
-print(f"hit-heavy: {ROUNDS * BATCH / dt / 1e6:.2f} M reads/s "
-      f"({dt / ROUNDS * 1e3:.2f} ms/batch of {BATCH})")
 
-t0 = time.perf_counter()
-for _ in range(ROUNDS):
-    db.multiget(miss, val_stride=128)
-dt = time.perf_counter() - t0
-print(f"all-miss:  {ROUNDS * len(miss) / dt / 1e6:.2f} M reads/s")
+def prebuilt(qs, stride=128):
+    kb = b"".join(qs)
+    refs = (ra.ffi.GraKeyRef * len(qs))()
+    o = 0
+    for j, k in enumerate(qs):
+        refs[j].off, refs[j].len = o, len(k)
+        o += len(k)
+    vals = C.create_string_buffer(len(qs) * stride)
+    outs = (ra.ffi.GraGetResult * len(qs))()
+    return kb, refs, vals, outs, stride
+
+
+def run(label, qs, expect_found):
+    kb, refs, vals, outs, stride = prebuilt(qs)
+    lib.gra_multiget(db.h, len(qs), refs, kb, len(kb), vals, stride, outs)
+    nf = sum(1 for j in range(len(qs)) if outs[j].status == 0)
+    assert (nf == len(qs)) == expect_found, (label, nf)
+    t0 = time.perf_counter()
+    for _ in range(ROUNDS):
+        rc = lib.gra_multiget(db.h, len(qs), refs, kb, len(kb), vals, stride,
+                              outs)
+        assert rc == 0
+    dt = time.perf_counter() - t0
+    print(f"{label}: {ROUNDS * len(qs) / dt / 1e6:.2f} M reads/s "
+          f"({dt / ROUNDS * 1e3:.3f} ms/batch of {len(qs)})")
+
+
+run("hit-heavy (C-ABI)", probe, True)
+run("all-miss  (C-ABI)", miss, False)
+run("hit-64k   (C-ABI)", [keys[rng.randrange(N_ENTRIES)]
+                          for _ in range(65536)], True)
 e.close()
