@@ -315,6 +315,11 @@ def main():
                 "copy": st.copy_ms / max(st.ticks, 1),
                 "total": st.total_ms / max(st.ticks, 1),
             },
+            # secondary leg by design: the exchange step of
+            # config #4. The headline line (no --repartition) carries the
+            # roofline and cpu_baseline objects; this one measures the
+            # RCCL-all-to-all + zero-copy apply path.
+            "secondary": True,
         }
         if rank == 0:
             print(json.dumps(result))
