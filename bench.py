@@ -50,6 +50,8 @@ def parse_args():
                    help="optional rocprofv3-derived per-launch HBM traffic (see profiles/)")
     p.add_argument("--h2d", action="store_true",
                    help="measure the PCIe-inclusive staging path instead (side report)")
+    p.add_argument("--no-legs", action="store_true",
+                   help="skip the h2d/drain side legs (clean-profile runs)")
     p.add_argument("--snappy", action="store_true",
                    help="BASELINE config #5: Snappy-compressed payloads, "
                         "GPU decompress stage ahead of the decode walk")
@@ -491,7 +493,8 @@ def main():
                       "records": st.records},
     }
 
-    if rank == 0 and world == 1 and not args.h2d and not args.snappy:
+    if (rank == 0 and world == 1 and not args.h2d and not args.snappy
+            and not args.no_legs):
         # staging-inclusive side line on every headline run
         result["h2d_staged"] = h2d_side_leg(eng, arena, used, descs, n_upd, args)
         # host-drain side line; separate engine so the
