@@ -9,7 +9,7 @@ cd /tmp && export TMPDIR=/tmp
 REPO=${GRAFT_REPO_ROOT:-/root/repo}
 OUT=$REPO/gpurun_out/prof
 mkdir -p "$OUT"
-BENCH="python $REPO/bench.py --steps 6 --warmup 2 --max-ticks-resident 4 --no-cpu-baseline --no-legs"
+BENCH="python $REPO/bench.py --steps 24 --warmup 6 --max-ticks-resident 4 --no-cpu-baseline --no-legs"
 SNAPPY_BENCH="python $REPO/bench.py --snappy --kind 2 --steps 4 --warmup 2 --max-ticks-resident 2 --no-cpu-baseline"
 
 rocprofv3 --output-format csv --kernel-trace --stats -d "$OUT/stats" -o run -- $BENCH > "$OUT/bench_stats.json" 2> "$OUT/stats.log"
