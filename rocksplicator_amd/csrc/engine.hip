@@ -283,9 +283,20 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = btag;
     h.flags = cf4 ? 1 : 0;
-    h.kpref = 0; /* k_copy fills the key fingerprint from the bytes it is
-                    streaming anyway; lets k_multiget reject non-matching
-                    entries on the header alone, no payload touch */
+    /* kpref = FNV fingerprint of the STORED key ([cf LE4|key] for cf
+     * records): header-only rejects in k_multiget. Filled HERE from the
+     * blob key bytes (+~35 us/tick measured) — a k_copy-side fill was 13x
+     * costlier (scattered lane-0 read-modify-writes of header lines,
+     * ~480 us/launch; profiles/r02). */
+    {
+      uint32_t fh = wb::kFnvBasis32;
+      if (cf4) {
+        uint8_t cfb[4] = {(uint8_t)r.cf_id, (uint8_t)(r.cf_id >> 8),
+                          (uint8_t)(r.cf_id >> 16), (uint8_t)(r.cf_id >> 24)};
+        fh = wb::key_fnv_fold(fh, cfb, 4);
+      }
+      h.kpref = wb::key_fnv_fold(fh, blobs + d.off + r.key_off, r.key_len);
+    }
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -293,9 +304,7 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     CopyTask tk;
     tk.src_off = d.off + r.key_off;
     tk.dst_rel = pay + cf4;
-    /* bit31 marks a cf key task: k_copy must NOT overwrite the kpref emit
-     * just set (the stored prefix is the cf id, not the raw key bytes) */
-    tk.nbytes = r.key_len | (cf4 ? 0x80000000u : 0u);
+    tk.nbytes = r.key_len;
     tasks[2 * (rec + idx)] = tk;
     if (cfv) { /* unaligned position: byte stores */
       uint8_t *p = pay_region + pay + cf4 + r.key_len;
@@ -389,22 +398,10 @@ __global__ void __launch_bounds__(256) k_copy(const uint8_t *__restrict__ blobs,
   uint32_t g = (blockIdx.x * blockDim.x + threadIdx.x) / G;
   uint32_t ngroups = (gridDim.x * blockDim.x) / G;
   uint8_t *pay_region = store + place->payload_off;
-  wb::RecHdr *hdrs = (wb::RecHdr *)(store + place->hdr_off);
   for (uint32_t t = g; t < ntasks; t += ngroups) {
     CopyTask tk = tasks[t];
-    uint32_t nb = tk.nbytes & 0x7FFFFFFFu;
-    if ((t & 1) == 0 && lane == 0) {
-      /* key task: fingerprint the STORED key from bytes this group is
-       * about to stream anyway (same cachelines — near free; the r01
-       * emit-side attempt re-touched COLD blob bytes and lost). For cf
-       * records fold the 4 prefix bytes k_emit just wrote first. */
-      uint32_t fh = wb::kFnvBasis32;
-      if (tk.nbytes >> 31)
-        fh = wb::key_fnv_fold(fh, pay_region + tk.dst_rel - 4, 4);
-      hdrs[t >> 1].kpref = wb::key_fnv_fold(fh, blobs + tk.src_off, nb);
-    }
-    if (nb == 0) continue;
-    copy_dwordx4<G>(pay_region + tk.dst_rel, blobs + tk.src_off, nb, lane);
+    if (tk.nbytes == 0) continue;
+    copy_dwordx4<G>(pay_region + tk.dst_rel, blobs + tk.src_off, tk.nbytes, lane);
   }
 }
 
