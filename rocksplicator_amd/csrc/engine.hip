@@ -283,20 +283,11 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = btag;
     h.flags = cf4 ? 1 : 0;
-    /* kpref = FNV fingerprint of the STORED key ([cf LE4|key] for cf
-     * records): header-only rejects in k_multiget. Filled HERE from the
-     * blob key bytes (+~35 us/tick measured) — a k_copy-side fill was 13x
-     * costlier (scattered lane-0 read-modify-writes of header lines,
-     * ~480 us/launch; profiles/r02). */
-    {
-      uint32_t fh = wb::kFnvBasis32;
-      if (cf4) {
-        uint8_t cfb[4] = {(uint8_t)r.cf_id, (uint8_t)(r.cf_id >> 8),
-                          (uint8_t)(r.cf_id >> 16), (uint8_t)(r.cf_id >> 24)};
-        fh = wb::key_fnv_fold(fh, cfb, 4);
-      }
-      h.kpref = wb::key_fnv_fold(fh, blobs + d.off + r.key_off, r.key_len);
-    }
+    /* kpref computed during the walk (wb::Rec.kpref): the key's first
+     * cacheline is hot there from the length-varint read. Alternatives
+     * measured and rejected: k_copy-side scattered header RMWs
+     * (+480 us/launch), emit-side cold gather (+158 us) — profiles/r02. */
+    h.kpref = r.kpref;
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;

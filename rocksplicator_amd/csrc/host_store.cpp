@@ -157,12 +157,7 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
-      /* same stored-key fingerprint the GPU copy kernel fills */
-      {
-        uint32_t fh = wb::kFnvBasis32;
-        if (cf4) fh = wb::key_fnv_fold(fh, (const uint8_t *)&rc.cf_id, 4);
-        h.kpref = wb::key_fnv_fold(fh, rep + rc.key_off, rc.key_len);
-      }
+      h.kpref = rc.kpref; /* fingerprint computed during the walk */
       hd[i] = h;
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);
