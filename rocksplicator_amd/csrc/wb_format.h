@@ -81,6 +81,17 @@ WB_HD uint32_t varint32(const uint8_t *p, uint32_t avail, uint32_t *out) {
   return 0;
 }
 
+/* 32-bit FNV-1a fingerprint of the STORED key bytes ([cf LE4 | key] for cf
+ * records) — the kpref filter value. A raw prefix was useless for real key
+ * families (shared "user..."-style prefixes reject nothing); a hash keeps
+ * ~2^-32 false-accept regardless of key structure. Seeded fold helper so
+ * the cf prefix can be folded first. */
+constexpr uint32_t kFnvBasis32 = 2166136261u;
+WB_HD uint32_t key_fnv_fold(uint32_t h, const uint8_t *p, uint32_t n) {
+  for (uint32_t i = 0; i < n; i++) h = (h ^ p[i]) * 16777619u;
+  return h;
+}
+
 /* Does this record tag consume a sequence number?
  * Put/Delete/SingleDelete/Merge/RangeDeletion (+CF variants) do; LogData,
  * Noop and 2PC markers do not (assumption test :136-187). */
@@ -247,17 +258,6 @@ WB_HD uint32_t key_prefix4(const uint8_t *key, uint32_t klen) {
   uint32_t p = 0;
   for (uint32_t i = 0; i < 4 && i < klen; i++) p |= (uint32_t)key[i] << (8 * i);
   return p;
-}
-
-/* 32-bit FNV-1a fingerprint of the STORED key bytes ([cf LE4 | key] for cf
- * records) — the kpref filter value. A raw prefix was useless for real key
- * families (shared "user..."-style prefixes reject nothing); a hash keeps
- * ~2^-32 false-accept regardless of key structure. Seeded fold helper so
- * the cf prefix can be folded first. */
-constexpr uint32_t kFnvBasis32 = 2166136261u;
-WB_HD uint32_t key_fnv_fold(uint32_t h, const uint8_t *p, uint32_t n) {
-  for (uint32_t i = 0; i < n; i++) h = (h ^ p[i]) * 16777619u;
-  return h;
 }
 static_assert(sizeof(RecHdr) == 24, "RecHdr must be 24 bytes");
 
