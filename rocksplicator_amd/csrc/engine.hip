@@ -283,11 +283,13 @@ __global__ void k_emit(const uint8_t *__restrict__ blobs,
     h.key_len = (uint16_t)(r.key_len + cf4);
     h.type = btag;
     h.flags = cf4 ? 1 : 0;
-    /* kpref computed during the walk (wb::Rec.kpref): the key's first
-     * cacheline is hot there from the length-varint read. Alternatives
-     * measured and rejected: k_copy-side scattered header RMWs
-     * (+480 us/launch), emit-side cold gather (+158 us) — profiles/r02. */
-    h.kpref = r.kpref;
+    h.kpref = 0; /* the read index is built LAZILY at first multiget
+                    (k_kpref, one coalesced pass over the run) — every
+                    apply-time fill variant measured 160-480 us/tick
+                    against the headline (profiles/r02): k_copy scattered
+                    header RMWs +480, emit cold gather +158, walk-side
+                    fold +180 in k_decode. rocksdb's analog: filters are
+                    built at flush, not per write. */
     hdrs[rec + idx] = h;
     if (cf4) { /* record start is 16-B aligned -> u32 store is aligned */
       *(uint32_t *)(pay_region + pay) = r.cf_id;
@@ -591,6 +593,26 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
   if (tid == 0) {
     out[q].status = GRA_GET_FOUND;
     out[q].vlen = h.val_len; /* true length (caller sees truncation) */
+  }
+}
+
+/* Lazy read-index build: one coalesced pass over a run fills every
+ * header's key fingerprint from the STORED key bytes (contiguous in the
+ * payload). Launched at a run's first multiget — apply-time fills all
+ * measured 160-480 us/tick against the headline (see k_emit). */
+__global__ void k_kpref(uint8_t *__restrict__ store,
+                        const RunView *__restrict__ runs) {
+  RunView rv = runs[blockIdx.y];
+  wb::RecHdr *hdrs = (wb::RecHdr *)(store + rv.hdr_off);
+  const uint8_t *pay = store + rv.payload_off;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < rv.n_entries;
+       i += gridDim.x * blockDim.x) {
+    wb::RecHdr h = hdrs[i];
+    hdrs[i].kpref =
+        (h.type == wb::kRangeDeletion)
+            ? 0 /* tombstones are range-matched, not hash-matched */
+            : wb::key_fnv_fold(wb::kFnvBasis32,
+                               pay + (h.kv_off - rv.pay_rel_base), h.key_len);
   }
 }
 
@@ -2117,7 +2139,7 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
    * two halves merge by seq — a mixed shard no longer forces the whole
    * call to the host path */
   std::vector<RunView> views;
-  std::vector<std::shared_ptr<Run>> host_runs;
+  std::vector<std::shared_ptr<Run>> host_runs, unbuilt;
   {
     std::lock_guard<std::mutex> lk(ss.mu);
     views.reserve(ss.runs.size());
@@ -2129,6 +2151,9 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
                                      tracks max seqs */
       } else {
         views.push_back({r.hdr_cur, r.payload_cur, r.n_entries, r.pay_rel_base});
+        if (!r.kpref_built) unbuilt.push_back(*it); /* racy pre-filter:
+                                     re-checked under e->mu (rebuilds are
+                                     idempotent anyway) */
       }
     }
   }
@@ -2192,6 +2217,41 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
       g_err = "gra_multiget: allocation failed";
       return GRA_ERR;
     }
+  }
+  /* lazy read-index build for first-served runs (single pass per run,
+   * ordered before this call's lookup kernels on the same stream;
+   * concurrent calls serialize on e->mu and re-check the flag) */
+  std::vector<RunView> bviews;
+  for (auto &rp : unbuilt) {
+    if (!rp->kpref_built) {
+      bviews.push_back(
+          {rp->hdr_cur, rp->payload_cur, rp->n_entries, rp->pay_rel_base});
+      rp->kpref_built = true;
+    }
+  }
+  if (!bviews.empty()) {
+    size_t need = (views.size() > bviews.size() ? views.size()
+                                                : bviews.size()) *
+                  sizeof(RunView);
+    if (!grow(&mg.d_runs, &mg.runs_cap, need) ||
+        hipMemcpyAsync(mg.d_runs, bviews.data(),
+                       bviews.size() * sizeof(RunView),
+                       hipMemcpyHostToDevice, e->stream) != hipSuccess) {
+      g_err = "gra_multiget: index build failed";
+      return GRA_ERR;
+    }
+    uint64_t maxe = 0;
+    for (auto &v : bviews)
+      maxe = v.n_entries > maxe ? v.n_entries : maxe;
+    uint32_t bx = (uint32_t)((maxe + 255) / 256);
+    if (bx > 256) bx = 256;
+    hipLaunchKernelGGL(k_kpref, dim3(bx, (uint32_t)bviews.size()), dim3(256),
+                       0, e->stream, e->d_store, mg.d_runs);
+    if (hipGetLastError() != hipSuccess) {
+      g_err = "gra_multiget: index build launch failed";
+      return GRA_ERR;
+    }
+    d_runs = mg.d_runs; /* grow may have reallocated */
   }
   std::vector<uint64_t> qtab;
   if (hashjoin) { /* host-built open-addressed fingerprint table */

@@ -157,11 +157,14 @@ bool host_build_run(const uint8_t *rep, size_t len, uint64_t base_seq, Run *out)
       h.key_len = (uint16_t)(rc.key_len + cf4);
       h.type = wb::base_tag(rc.tag);
       h.flags = cf4 ? 1 : 0;
-      h.kpref = rc.kpref; /* fingerprint computed during the walk */
       hd[i] = h;
       uint8_t *p = out->payload.data() + off;
       if (cf4) memcpy(p, &rc.cf_id, 4);
       memcpy(p + cf4, rep + rc.key_off, rc.key_len);
+      /* fingerprint over the STORED key bytes just written (host runs pay
+       * it eagerly — CPU cost is trivial; device runs build theirs lazily
+       * at first multiget via k_kpref) */
+      hd[i].kpref = wb::key_fnv_fold(wb::kFnvBasis32, p, cf4 + rc.key_len);
       if (cfv) memcpy(p + cf4 + rc.key_len, &rc.cf_id, 4);
       memcpy(p + cf4 + rc.key_len + cfv, rep + rc.val_off, rc.val_len);
       off += (cf4 + rc.key_len + cfv + rc.val_len + 15u) & ~15u;

@@ -43,6 +43,8 @@ struct Run {
    * points at the tick's payload base so pay_p + kv_off is correct. */
   std::shared_ptr<uint8_t> hbuf;
   const uint8_t *hdr_p = nullptr, *pay_p = nullptr;
+  bool kpref_built = false; /* device read index (k_kpref) done; guarded by
+                               the engine mutex at build time */
   const uint8_t *hdrs_data() const { return hdr_p ? hdr_p : hdrs.data(); }
   const uint8_t *payload_data() const { return pay_p ? pay_p : payload.data(); }
   bool resident() const {

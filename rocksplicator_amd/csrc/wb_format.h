@@ -139,9 +139,6 @@ struct Rec {
   uint8_t consumes;  /* 1 if seq-consuming */
   uint16_t _pad;
   uint32_t cf_id;
-  uint32_t kpref;    /* FNV fingerprint of the stored key ([cfLE4|key]) —
-                        computed during the walk while the key's first
-                        cacheline is hot from the length-varint read */
 };
 
 /* Walk result for one blob. */
@@ -199,15 +196,6 @@ WB_HD WalkTotals walk_f(const uint8_t *rep, uint32_t len, F &&f) {
       r.consumes = 1;
       r._pad = 0;
       r.cf_id = cf;
-      {
-        uint32_t fh = kFnvBasis32;
-        if (cf) {
-          uint8_t cfb[4] = {(uint8_t)cf, (uint8_t)(cf >> 8),
-                            (uint8_t)(cf >> 16), (uint8_t)(cf >> 24)};
-          fh = key_fnv_fold(fh, cfb, 4);
-        }
-        r.kpref = key_fnv_fold(fh, rep + off[0], slen[0]);
-      }
       f(r, nrec);
       /* payload stored per record: [cf_id?4B] key [cf_id?4B-for-range-end]
        * value, 16-B aligned. CF range tombstones prefix BOTH slices (begin
