@@ -520,3 +520,48 @@ def test_rejected_batches_never_served_downstream(olib):
     ups = mdb.get_updates(0)
     assert [u[0] for u in ups] == [1, 2]
     mid.close()
+
+
+def test_thrift_wire_replication(olib):
+    """The SAME pull exchange over the reference's own wire format:
+    THeader framing + compact-protocol Replicator.replicate
+    (thrift_wire.py restates fbthrift HeaderClientChannel's default wire
+    per the published specs; replicator.thrift:21-90 structs). A follower
+    on this framework byte-speaks the reference protocol."""
+    from rocksplicator_amd import thrift_wire as tw
+    from rocksplicator_amd.replicator import pull_once
+    nshards = 3
+    leader = ra.Engine(nshards=nshards, merge_op=1, retain_log=1)
+    follower = ra.Engine(nshards=nshards, merge_op=1)
+    srv = tw.ThriftUpdateServer()
+    ldbs, fdbs, remotes = [], [], []
+    for s in range(nshards):
+        ldbs.append(leader.open(s))
+        fdbs.append(follower.open(s))
+        srv.register(f"db{s}", ldbs[s])
+        remotes.append(tw.ThriftRemoteUpstream("127.0.0.1", srv.port,
+                                               f"db{s}"))
+    rng = random.Random(31)
+    one = (1).to_bytes(8, "little")
+    for i in range(300):
+        s = rng.randrange(nshards)
+        rep = (PyBatch().merge(b"ctr", one).data() if rng.random() < 0.5
+               else PyBatch().put(f"k{rng.randrange(30)}".encode(),
+                                  rng.randbytes(64)).data())
+        ldbs[s].write_leader(rep)
+        srv.notify_write()
+    for s in range(nshards):
+        while pull_once(remotes[s], fdbs[s]):
+            pass
+        follower.flush()
+        assert fdbs[s].latest_seq() == ldbs[s].latest_seq()
+        assert fdbs[s].get(b"ctr") == ldbs[s].get(b"ctr")
+        for i in range(30):
+            k = f"k{i}".encode()
+            assert fdbs[s].get(k) == ldbs[s].get(k), (s, k)
+        assert fdbs[s].checksum() == ldbs[s].checksum()
+    for r in remotes:
+        r.close()
+    srv.close()
+    leader.close()
+    follower.close()
