@@ -846,7 +846,8 @@ struct Stats {
 };
 
 constexpr int kSlots = 8;
-constexpr int kEventsPerTick = 10;
+constexpr int kEventsPerTick = 12; /* 10 = decode start (prep stream),
+                                      11 = decode done */
 
 struct TickRec {
   int slot = -1;
@@ -905,9 +906,16 @@ struct GraEngine {
   uint32_t max_upd;          /* max updates per tick */
   uint32_t task_cap;
   uint32_t group_cap;        /* max contiguous shard-groups per tick */
-  wb::WalkTotals *d_totals = nullptr;
-  uint2 *d_partial = nullptr, *d_bsums = nullptr;
-  wb::Rec *d_reccache = nullptr;
+  /* decode scratch is PARITY-DOUBLED: tick N+1's k_decode runs on the
+   * prep stream under tick N's k_copy (decode is a sparse latency-bound
+   * walk, copy is HBM-bound — they compose), so consecutive ticks must
+   * not share totals/scans/reccache. scratch_used_ev[p] (recorded after
+   * a tick's last scratch reader, k_rundesc) gates reuse. */
+  wb::WalkTotals *d_totals_b[2] = {nullptr, nullptr};
+  uint2 *d_partial_b[2] = {nullptr, nullptr};
+  uint2 *d_bsums_b[2] = {nullptr, nullptr};
+  wb::Rec *d_reccache_b[2] = {nullptr, nullptr};
+  hipEvent_t scratch_used_ev[2] = {nullptr, nullptr};
   CopyTask *d_tasks = nullptr;
   TickPlace *d_place = nullptr; /* kSlots entries: per-tick placement slot
                                    (the drain D2H snapshot on copyout must
@@ -1122,10 +1130,14 @@ int GraEngine::init(const GraEngineOpts &o) {
   HIP_TRY(hipMalloc(&d_store, opts.store_bytes + 16));
   HIP_TRY(hipMalloc(&d_cursor, 8));
   HIP_TRY(hipMemset(d_cursor, 0, 8));
-  HIP_TRY(hipMalloc(&d_totals, (size_t)max_upd * sizeof(wb::WalkTotals)));
-  HIP_TRY(hipMalloc(&d_partial, (size_t)max_upd * sizeof(uint2)));
-  HIP_TRY(hipMalloc(&d_bsums, ((size_t)max_upd / 256 + 2) * sizeof(uint2)));
-  HIP_TRY(hipMalloc(&d_reccache, (size_t)max_upd * 2 * sizeof(wb::Rec)));
+  for (int i = 0; i < 2; i++) {
+    HIP_TRY(hipMalloc(&d_totals_b[i], (size_t)max_upd * sizeof(wb::WalkTotals)));
+    HIP_TRY(hipMalloc(&d_partial_b[i], (size_t)max_upd * sizeof(uint2)));
+    HIP_TRY(hipMalloc(&d_bsums_b[i], ((size_t)max_upd / 256 + 2) * sizeof(uint2)));
+    HIP_TRY(hipMalloc(&d_reccache_b[i], (size_t)max_upd * 2 * sizeof(wb::Rec)));
+    HIP_TRY(hipEventCreateWithFlags(&scratch_used_ev[i], hipEventDisableTiming));
+    HIP_TRY(hipEventRecord(scratch_used_ev[i], stream));
+  }
   HIP_TRY(hipMalloc(&d_tasks, (size_t)task_cap * sizeof(CopyTask)));
   HIP_TRY(hipMalloc(&d_place, kSlots * sizeof(TickPlace)));
   HIP_TRY(hipMalloc(&d_groups, (size_t)group_cap * sizeof(GroupDesc)));
@@ -1217,10 +1229,14 @@ GraEngine::~GraEngine() {
     if (stage_used_ev[i]) (void)hipEventDestroy(stage_used_ev[i]);
     if (h2d_gate_ev[i]) (void)hipEventDestroy(h2d_gate_ev[i]);
   }
-  for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_totals,
-                  (void *)d_partial, (void *)d_bsums, (void *)d_reccache,
-                  (void *)d_tasks, (void *)d_place, (void *)d_groups,
-                  (void *)d_err_ring})
+  for (int i = 0; i < 2; i++) {
+    for (void *p : {(void *)d_totals_b[i], (void *)d_partial_b[i],
+                    (void *)d_bsums_b[i], (void *)d_reccache_b[i]})
+      if (p) (void)hipFree(p);
+    if (scratch_used_ev[i]) (void)hipEventDestroy(scratch_used_ev[i]);
+  }
+  for (void *p : {(void *)d_store, (void *)d_cursor, (void *)d_tasks,
+                  (void *)d_place, (void *)d_groups, (void *)d_err_ring})
     if (p) (void)hipFree(p);
   shards.clear(); /* release run arenas before draining the pool */
   pending.clear();
@@ -1362,20 +1378,50 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
     if (snappy_overlap)
       HIP_TRY(hipStreamWaitEvent(stream, t.ev[8], 0)); /* decode gates on it */
   }
-  hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
-                     n, d_totals, opts.max_wb_records, d_err_ring, tick,
-                     d_partial, d_bsums, d_reccache, sl.d_ok);
+  /* decode runs on the PREP stream, overlapped with the PREVIOUS tick's
+   * emit/copy on main (decode is a sparse latency-bound walk over record
+   * headers; copy is HBM-bound — they compose). Parity scratch + the
+   * scratch_used gate make consecutive ticks independent; main's scan2
+   * gates on decode-done (ev11). GRA_DECODE_OVERLAP=0 restores the serial
+   * order for A/B. */
+  static const bool decode_overlap = [] {
+    const char *v = getenv("GRA_DECODE_OVERLAP");
+    return !v || v[0] != '0';
+  }();
+  int par = (int)(tick & 1u);
+  wb::WalkTotals *t_totals = d_totals_b[par];
+  uint2 *t_partial = d_partial_b[par];
+  uint2 *t_bsums = d_bsums_b[par];
+  wb::Rec *t_reccache = d_reccache_b[par];
+  hipStream_t ds = decode_overlap ? prep : stream;
+  if (decode_overlap) {
+    /* previous user of this parity's scratch must have finished reading */
+    HIP_TRY(hipStreamWaitEvent(prep, scratch_used_ev[par], 0));
+    if (stage_buf >= 0) /* staged blobs land via the h2d stream */
+      HIP_TRY(hipStreamWaitEvent(prep, t.ev[1], 0));
+    t.evmask |= 1u << 10;
+    HIP_TRY(hipEventRecord(t.ev[10], prep));
+  }
+  hipLaunchKernelGGL(k_decode, dim3(nb), dim3(256), 0, ds, d_blobs, d_descw,
+                     n, t_totals, opts.max_wb_records, d_err_ring, tick,
+                     t_partial, t_bsums, t_reccache, sl.d_ok);
   HIP_TRY(hipGetLastError());
-  if (detailed) HIP_TRY(rec(2)); /* after decode(+scan1) */
+  if (decode_overlap) {
+    t.evmask |= 1u << 11;
+    HIP_TRY(hipEventRecord(t.ev[11], prep));
+    HIP_TRY(hipStreamWaitEvent(stream, t.ev[11], 0));
+  } else if (detailed) {
+    HIP_TRY(rec(2)); /* after decode(+scan1) */
+  }
   TickPlace *place_slot = d_place + si;
-  hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, d_bsums, nb,
+  hipLaunchKernelGGL(k_scan2, dim3(1), dim3(256), 0, stream, t_bsums, nb,
                      d_cursor, place_slot, opts.store_bytes, opts.store_ring,
                      task_cap, d_err_ring, tick);
   HIP_TRY(hipGetLastError());
   if (detailed) HIP_TRY(rec(3)); /* after scan(+reserve) */
   hipLaunchKernelGGL(k_emit, dim3(nb), dim3(256), 0, stream, d_blobs, d_descw,
-                     n, d_totals, d_partial, d_bsums, place_slot, d_store,
-                     d_tasks, d_reccache);
+                     n, t_totals, t_partial, t_bsums, place_slot, d_store,
+                     d_tasks, t_reccache);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(4)); /* after emit (pre-copy) */
   /* group width by average update size (micro_copy.hip: g32 wins >=512B) */
@@ -1391,10 +1437,11 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   if (window_ev) /* this window's scratch fully consumed */
     HIP_TRY(hipEventRecord(window_ev, stream));
   hipLaunchKernelGGL(k_rundesc, dim3((ngroups + 255) / 256), dim3(256), 0,
-                     stream, groups_for_kernel, ngroups, d_descw, d_totals,
-                     d_partial, d_bsums, n, nb, place_slot, sl.d_rundescs);
+                     stream, groups_for_kernel, ngroups, d_descw, t_totals,
+                     t_partial, t_bsums, n, nb, place_slot, sl.d_rundescs);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(6)); /* main-stream tick end */
+  HIP_TRY(hipEventRecord(scratch_used_ev[par], stream)); /* scratch free */
   if (stage_buf >= 0) /* last reader of this staging buffer has retired */
     HIP_TRY(hipEventRecord(stage_used_ev[stage_buf], stream));
   /* publish run descriptors on the copyout stream, overlapped with the next
@@ -1466,7 +1513,11 @@ int GraEngine::ingest_one(TickRec &t, bool wait) {
      * (decode..copy) is unaffected by it */
     stats.snappy_ms += dt(has(9) ? 9 : prev, 8);
   }
-  if (has(2)) {
+  if (has(10) && has(11)) {
+    /* decode overlapped on the prep stream: its own duration, main chain
+     * unaffected */
+    stats.decode_ms += dt(10, 11);
+  } else if (has(2)) {
     stats.decode_ms += dt(prev, 2);
     prev = 2;
   }
