@@ -1424,13 +1424,21 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
                      d_tasks, t_reccache);
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(4)); /* after emit (pre-copy) */
-  /* group width by average update size (micro_copy.hip: g32 wins >=512B) */
+  /* group width by average update size (micro_copy.hip: g32 wins >=512B).
+   * Grid: 2048 blocks = 8/CU = every wave slot — the next tick's decode
+   * can only overlap if the copy leaves wave headroom, hence the
+   * GRA_COPY_GRID knob for the A/B (profiles/r02). */
+  static const uint32_t copy_grid = [] {
+    const char *v = getenv("GRA_COPY_GRID");
+    uint32_t g = v ? (uint32_t)atoi(v) : 2048;
+    return g ? g : 2048;
+  }();
   if (blob_bytes / (n ? n : 1) >= 512) {
-    hipLaunchKernelGGL((k_copy<32>), dim3(2048), dim3(256), 0, stream, d_blobs,
-                       d_store, place_slot, d_tasks);
+    hipLaunchKernelGGL((k_copy<32>), dim3(copy_grid), dim3(256), 0, stream,
+                       d_blobs, d_store, place_slot, d_tasks);
   } else {
-    hipLaunchKernelGGL((k_copy<16>), dim3(2048), dim3(256), 0, stream, d_blobs,
-                       d_store, place_slot, d_tasks);
+    hipLaunchKernelGGL((k_copy<16>), dim3(copy_grid), dim3(256), 0, stream,
+                       d_blobs, d_store, place_slot, d_tasks);
   }
   HIP_TRY(hipGetLastError());
   HIP_TRY(rec(5)); /* after copy */
