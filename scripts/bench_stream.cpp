@@ -22,12 +22,16 @@ int main(int argc, char **argv) {
   int nthreads = argc > 2 ? atoi(argv[2]) : 16;
   uint32_t vlen = argc > 3 ? atoi(argv[3]) : 1024;
   double seconds = argc > 4 ? atof(argv[4]) : 5.0;
+  uint32_t staging_mb = argc > 5 ? atoi(argv[5]) : 0; /* 0 = engine default */
+  int64_t ts = argc > 6 ? atoll(argv[6]) : 1; /* 0 disables the per-call
+                                                 latency clock sample */
 
   GraEngineOpts opts;
   gra_engine_opts_init(&opts);
   opts.nshards = nshards;
   opts.store_ring = 1;
   opts.store_bytes = 8ULL << 30;
+  if (staging_mb) opts.staging_bytes = (uint64_t)staging_mb << 20;
   GraEngine *e = nullptr;
   if (gra_engine_create(&opts, &e) != GRA_OK) {
     fprintf(stderr, "engine: %s\n", gra_last_error());
@@ -55,7 +59,7 @@ int main(int argc, char **argv) {
       uint64_t n = 0;
       while (!stop.load(std::memory_order_relaxed)) {
         for (GraDb *db : dbs) {
-          if (!gra_handle_replicate_response(db, bdata, blen, 1)) {
+          if (!gra_handle_replicate_response(db, bdata, blen, ts)) {
             fprintf(stderr, "apply refused\n");
             stop = true;
             break;
