@@ -1166,8 +1166,19 @@ int GraEngine::init(const GraEngineOpts &o) {
    * chunk could otherwise match a freshly created engine reusing the same
    * heap address (ABA) */
   static std::atomic<uint64_t> g_epoch{1};
+  /* staging pinned memory is written by MANY host threads and only read
+   * by SDMA H2D: allocate it NON-COHERENT (cacheable on the CPU side) —
+   * the default fine-grained pinned memory takes uncached CPU writes on
+   * these hosts and capped the streaming fill around 20 GB/s.
+   * GRA_STAGE_COHERENT=1 restores the default for A/B. */
+  static const unsigned stage_flags = [] {
+    const char *v = getenv("GRA_STAGE_COHERENT");
+    return (v && v[0] == '1') ? hipHostMallocDefault
+                              : hipHostMallocNonCoherent;
+  }();
   for (int i = 0; i < 2; i++) {
-    HIP_TRY(hipHostMalloc(&stage[i].pin, opts.staging_bytes + 16));
+    HIP_TRY(hipHostMalloc(&stage[i].pin, opts.staging_bytes + 16,
+                          stage_flags));
     stage[i].descs = (GraUpdateDesc *)malloc((size_t)max_upd * sizeof(GraUpdateDesc));
     stage[i].epoch = g_epoch.fetch_add(1u << 20);
     HIP_TRY(hipEventCreate(&stage[i].free_ev));
