@@ -1994,11 +1994,18 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
           sb->descs[i].len = 0;
         ck.sb = nullptr;
         w.fetch_sub(1, std::memory_order_acq_rel);
-        std::lock_guard<std::mutex> lk(e->mu);
+        /* double-checked: when a buffer fills, every writer lands here —
+         * only the first may take e->mu and build the tick (~15 ms for a
+         * 1M-update buffer); the rest must NOT queue on the mutex behind
+         * it (the swap publishes the next buffer before the build, so
+         * they can continue writing immediately) */
         if (e->cur_stage.load(std::memory_order_acquire) == sb) {
-          if (e->stream_tick_locked() != GRA_OK) {
-            rollback();
-            return 0;
+          std::lock_guard<std::mutex> lk(e->mu);
+          if (e->cur_stage.load(std::memory_order_acquire) == sb) {
+            if (e->stream_tick_locked() != GRA_OK) {
+              rollback();
+              return 0;
+            }
           }
         }
         continue;
