@@ -1963,7 +1963,6 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
   };
   thread_local Chunk ck;
   constexpr uint64_t kChunkBytes = 256 << 10;
-  constexpr uint32_t kChunkSlots = 128;
   /* striped writer slot: each thread RMWs its own cacheline (see
    * WriterSlot); >kWriterSlots threads share slots benignly */
   static std::atomic<uint32_t> g_wslot{0};
@@ -1985,12 +1984,18 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
                     ck.used + len <= ck.cap && ck.slots_left > 0;
     if (!chunk_ok) {
       uint64_t want = len > kChunkBytes ? (uint64_t)len : kChunkBytes;
+      /* slots sized for THIS call's update size so neither resource
+       * exhausts far ahead of the other (a fixed slot count abandoned
+       * ~half the byte range at 1 KB updates, and the H2D copies the
+       * whole [0, pos) range including dead space) */
+      uint32_t cslots = (uint32_t)(want / (len ? len : 1)) + 8;
+      if (cslots > 4096) cslots = 4096;
       uint64_t off = sb->pos.fetch_add(want, std::memory_order_relaxed);
-      uint32_t slot = sb->nslots.fetch_add(kChunkSlots, std::memory_order_relaxed);
+      uint32_t slot = sb->nslots.fetch_add(cslots, std::memory_order_relaxed);
       if (off + want + 16 > e->opts.staging_bytes ||
-          slot + kChunkSlots > e->max_upd) {
+          slot + cslots > e->max_upd) {
         /* buffer full: abandon the reserved slots (len = 0) and tick */
-        for (uint32_t i = slot; i < slot + kChunkSlots && i < e->max_upd; i++)
+        for (uint32_t i = slot; i < slot + cslots && i < e->max_upd; i++)
           sb->descs[i].len = 0;
         ck.sb = nullptr;
         w.fetch_sub(1, std::memory_order_acq_rel);
@@ -2010,8 +2015,8 @@ int gra_handle_replicate_response(GraDb *db, const uint8_t *rep, size_t len,
         }
         continue;
       }
-      for (uint32_t i = 0; i < kChunkSlots; i++) sb->descs[slot + i].len = 0;
-      ck = {e, sb, sb->epoch, off, 0, want, slot, kChunkSlots};
+      for (uint32_t i = 0; i < cslots; i++) sb->descs[slot + i].len = 0;
+      ck = {e, sb, sb->epoch, off, 0, want, slot, cslots};
     }
     uint64_t off = ck.base + ck.used;
     memcpy(sb->pin + off, rep, len);
