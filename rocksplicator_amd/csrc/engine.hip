@@ -1196,6 +1196,17 @@ int GraEngine::init(const GraEngineOpts &o) {
     HIP_TRY(hipHostMalloc(&s.h_err, 4));
     HIP_TRY(hipHostMalloc(&s.h_descs, (size_t)max_upd * sizeof(UpdDesc)));
   }
+  if (opts.drain_host && opts.store_ring &&
+      opts.store_bytes < 10 * opts.staging_bytes) {
+    /* k_drain reads a tick's store region on the copyout stream AFTER the
+     * main stream has moved on; a ring that wraps within the pipeline
+     * depth could overwrite it first. Enforce the safe sizing rather than
+     * risk a silent corruption (kSlots in-flight ticks + slack; a tick is
+     * bounded by staging_bytes of input). */
+    g_err = "drain_host with store_ring needs store_bytes >= 10x "
+            "staging_bytes (ring wrap inside the drain pipeline)";
+    return GRA_ERR;
+  }
   shards = std::vector<ShardState>(opts.nshards);
   /* pre-create the event pool for a full pipeline (hipEventCreate mid-run
    * showed up as ~1 ms hiccups in kernel traces) */
@@ -1298,6 +1309,13 @@ int GraEngine::enqueue_tick(const uint8_t *d_blobs, const UpdDesc *d_descw,
   if (n == 0) return GRA_OK;
   if (n > max_upd) {
     g_err = "tick exceeds max updates per tick";
+    return GRA_ERR;
+  }
+  if (opts.drain_host && opts.store_ring && blob_bytes * 10 > opts.store_bytes) {
+    /* same wrap hazard as the init-time guard, for replay ticks whose
+     * size isn't bounded by staging_bytes */
+    g_err = "drain_host with store_ring needs store_bytes >= 10x the tick "
+            "size (ring wrap inside the drain pipeline)";
     return GRA_ERR;
   }
   if (groups.size() > group_cap) {
