@@ -44,13 +44,29 @@ class ModelStore:
         self.tombs = []   # (seq, begin, end)
         self.latest = 0
 
+    @staticmethod
+    def _cfkey(cf, k):
+        import struct as _s
+        return _s.pack("<I", cf) + k
+
     def apply(self, ops):
+        """ops: (kind, a, b) with plain kinds, or cf kinds where a =
+        (cf_id, key) / (cf_id, begin, end) — cf-namespaced like the
+        engine/oracle: stored key = [cf LE4 | key], range tombstones
+        prefix BOTH bounds."""
         seq = self.latest + 1
         for kind, a, b in ops:
             if kind == "log_data":  # WAL-only marker: no seq, no memtable
                 continue
             if kind == "delete_range":
                 self.tombs.append((seq, a, b))
+            elif kind == "cf_delete_range":
+                cf, bk, ek = a
+                self.tombs.append((seq, self._cfkey(cf, bk),
+                                   self._cfkey(cf, ek)))
+            elif kind.startswith("cf_"):
+                cf, k = a
+                self.points.append((seq, kind[3:], self._cfkey(cf, k), b))
             else:
                 self.points.append((seq, kind, a, b))
             seq += 1
