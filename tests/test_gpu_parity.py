@@ -1043,3 +1043,69 @@ def test_multiget_mixed_host_device_runs(olib):
     for k, v in zip([b"a", b"b"], db.multiget([b"a", b"b"])):
         assert v == ost.get(0, k), k
     e.close()
+
+
+def test_multiget_hashjoin_overflow_fallback(olib):
+    """Many versions of few keys make the candidate list exceed its cap
+    (4*nq+1024): the hash-join must detect the overflow and retry via the
+    per-query scan kernel with identical results."""
+    e = ra.Engine(nshards=1)
+    db = e.open(0)
+    ost = oracle_ffi.Store(olib, 1)
+    # 3 keys x 2000 versions each -> any query of them yields ~2000
+    # candidates; 40 queries -> ~80000 >> 4*40+1024
+    for i in range(2000):
+        rep = (PyBatch().put(b"hot0", f"v{i}".encode())
+               .put(b"hot1", f"w{i}".encode())
+               .put(b"hot2", f"x{i}".encode()).data())
+        assert db.handle_replicate_response(rep)
+        assert ost.apply(0, rep)
+    e.flush()
+    probes = [f"hot{i % 3}".encode() for i in range(40)] + [b"cold"]
+    for k, v in zip(probes, db.multiget(probes)):
+        assert v == ost.get(0, k), k
+    e.close()
+
+
+def test_h2d_staged_corruption_recovery(olib):
+    """Corruption inside an h2d-staged window: the staged path carries no
+    per-update counts, so recovery must still poison the shard and roll
+    back to the durable boundary (whole-group drop)."""
+    e = ra.Engine(nshards=2)
+    dbs = [e.open(s) for s in range(2)]
+    good = [PyBatch().put(f"g{i}".encode(), b"v").data() for i in range(4)]
+    bad = bytearray(PyBatch().put(b"x", b"y").data())
+    bad[8] = 3
+    blobs = [(0, good[0]), (1, good[1]), (0, bytes(bad)), (0, good[2]),
+             (1, good[3])]
+    used = sum(len(b) for _s, b in blobs)
+    pin = e.pin_alloc(used + 64)
+    descs = (ra.ffi.GraUpdateDesc * len(blobs))()
+    off = 0
+    # shard-grouped order within the window (shard 0 first)
+    ordered = sorted(range(len(blobs)), key=lambda i: blobs[i][0])
+    base = C.cast(pin, C.c_void_p).value
+    for j, i in enumerate(ordered):
+        s, b = blobs[i]
+        C.memmove(base + off, b, len(b))
+        descs[j].shard, descs[j].len, descs[j].off, descs[j].ts = \
+            s, len(b), off, 0
+        off += len(b)
+    rep = e.upload(pin, used, descs, len(blobs))
+    rep.tick_h2d(0, len(blobs))
+    rep.sync()
+    # shard 0: g0 applied; bad drops the REST of the shard-0 group
+    # (no counts on the staged path -> whole-group truncation at the
+    # first corrupt update; g0 precedes it)
+    assert dbs[0].get(b"g0") == b"v"
+    assert dbs[0].get(b"g2") is None
+    assert dbs[0].latest_seq() == 1
+    # shard 1 untouched by shard 0's corruption
+    assert dbs[1].latest_seq() == 2
+    assert dbs[1].get(b"g1") == b"v" and dbs[1].get(b"g3") == b"v"
+    # fail-once then recovery on shard 0
+    assert not dbs[0].handle_replicate_response(good[2])
+    assert dbs[0].handle_replicate_response(good[2])
+    e.flush()
+    assert dbs[0].latest_seq() == 2 and dbs[0].get(b"g2") == b"v"
+    e.close()
