@@ -2864,9 +2864,12 @@ int gra_replay_tick_h2d(GraReplay *r, uint64_t first, uint64_t n) {
       return GRA_ERR;
     }
   }
+  std::vector<uint32_t> counts(r->counts.begin() + first,
+                               r->counts.begin() + first + n);
   return e->enqueue_tick(e->d_stage_blobs, plan.d_ud, (uint32_t)n,
                          plan.groups, plan.blob_bytes, true, r->h_arena + lo,
-                         hi - lo, e->d_stage_blobs, nullptr);
+                         hi - lo, e->d_stage_blobs, nullptr, plan.d_groups,
+                         nullptr, nullptr, std::move(counts));
 }
 
 int gra_replay_sync(GraReplay *r) { return gra_flush(r->e); }
