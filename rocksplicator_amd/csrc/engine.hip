@@ -2332,8 +2332,13 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
     return GRA_ERR;
   }
   if (!mg.d_counts) {
-    if (hipMalloc(&mg.d_counts, 8) != hipSuccess ||
-        hipHostMalloc(&mg.h_counts, 8) != hipSuccess) {
+    if (hipMalloc(&mg.d_counts, 8) != hipSuccess) {
+      g_err = "gra_multiget: allocation failed";
+      return GRA_ERR;
+    }
+    if (hipHostMalloc(&mg.h_counts, 8) != hipSuccess) {
+      (void)hipFree(mg.d_counts);
+      mg.d_counts = nullptr;
       g_err = "gra_multiget: allocation failed";
       return GRA_ERR;
     }
