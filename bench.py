@@ -52,6 +52,10 @@ def parse_args():
                    help="measure the PCIe-inclusive staging path instead (side report)")
     p.add_argument("--no-legs", action="store_true",
                    help="skip the h2d/drain side legs (clean-profile runs)")
+    p.add_argument("--device-wrap", action="store_true",
+                   help="TESTING ONLY: map rank devices modulo the visible "
+                        "GPU count so a multi-rank run can be validated on "
+                        "a 1-GPU box (never for real measurements)")
     p.add_argument("--snappy", action="store_true",
                    help="BASELINE config #5: Snappy-compressed payloads, "
                         "GPU decompress stage ahead of the decode walk")
@@ -346,6 +350,10 @@ def main():
     assert rc == 0, f"gen rc={rc}"
     used = used.value
     blob_bytes_total = used
+
+    if args.device_wrap:
+        import torch
+        local = local % max(1, torch.cuda.device_count())
 
     # --- engine + upload (untimed; inputs land in HBM) ---
     store_bytes = min(int(used * 1.3) + (1 << 30), 48 << 30)
