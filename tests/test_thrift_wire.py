@@ -175,3 +175,37 @@ def test_long_poll_wakeup():
         cli.close()
     finally:
         srv.close()
+
+
+def test_long_form_list_roundtrip():
+    """>=15 updates exercises the compact long-form list header
+    (0xF0|etype + varint size) — the shape every full pull response
+    (max_updates=50) uses."""
+    ups = [(i + 1, 10 * i, bytes([i]) * (i + 1)) for i in range(50)]
+    enc = tw.encode_replicate_response(ups, role=tw.ROLE_LEADER)
+    assert enc[1] == 0xF0 | tw.CT_STRUCT  # long-form list header
+    back = tw.decode_replicate_response(tw.Reader(enc))
+    assert [(u["seq_no"], u["timestamp"], u["raw_data"])
+            for u in back["updates"]] == ups
+
+
+def test_large_binary_and_field_long_form():
+    # 70 KB raw_data (3-byte varint length) + a long-form field id jump
+    big = bytes(range(256)) * 280
+    w = tw.StructWriter()
+    w.begin_struct()
+    w.binary(1, big)
+    w.i64(200, 7)  # delta > 15 -> long-form header with zigzag fid
+    w.end_struct()
+    got = {}
+
+    def f(fid, ct, rr):
+        if fid == 1:
+            got["b"] = rr.binary()
+        elif fid == 200:
+            got["x"] = rr.zz()
+        else:
+            tw.skip_field(rr, ct)
+
+    tw.read_struct_fields(tw.Reader(bytes(w.out)), f)
+    assert got == {"b": big, "x": 7}
