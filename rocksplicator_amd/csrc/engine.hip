@@ -2128,6 +2128,8 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
    * (replicated_db.cpp:452-456) */
   if (requester_role == 0 && since_seq > ss.acked_confirmed)
     ss.acked_confirmed = since_seq;
+  if (max_updates == 0) /* 0 = no limit (replicator.thrift:36-38) */
+    max_updates = UINT32_MAX;
   *n_out = 0;
   if (!ss.log.empty() && since_seq + 1 < ss.log.front().base_seq) {
     /* reference analog: WAL no longer reaches back that far */
