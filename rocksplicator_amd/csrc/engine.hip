@@ -2128,8 +2128,10 @@ int gra_get_updates(GraDb *db, uint64_t since_seq, uint32_t max_updates,
    * (replicated_db.cpp:452-456) */
   if (requester_role == 0 && since_seq > ss.acked_confirmed)
     ss.acked_confirmed = since_seq;
-  if (max_updates == 0) /* 0 = no limit (replicator.thrift:36-38) */
-    max_updates = UINT32_MAX;
+  /* NOTE: the C-ABI contract is that `out` holds max_updates entries, so
+   * max_updates==0 returns nothing HERE; the wire layers translate the
+   * IDL's 0-means-unlimited (replicator.thrift:36-38) into a bounded
+   * allocation before calling down (ffi.Db.get_updates). */
   *n_out = 0;
   if (!ss.log.empty() && since_seq + 1 < ss.log.front().base_seq) {
     /* reference analog: WAL no longer reaches back that far */

@@ -404,7 +404,11 @@ class Db:
                     observer=False):
         """Leader serving: [(seq, ts, rep_bytes), ...] with
         base seq > since_seq — the reference Update triple. An observer's
-        request does not post an ACK (replicated_db.cpp:452-456)."""
+        request does not post an ACK (replicated_db.cpp:452-456).
+        max_updates=0 means no limit (replicator.thrift:36-38), bounded
+        here by the byte cap via a generous entry allocation."""
+        if max_updates == 0:
+            max_updates = max(64, cap // 16)  # every entry needs >=12B rep
         out = (GraServedUpdate * max_updates)()
         buf = C.create_string_buffer(cap)
         n = C.c_uint32()
