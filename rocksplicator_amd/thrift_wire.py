@@ -472,8 +472,10 @@ class ThriftUpdateServer:
                         deadline = time.monotonic() + req["max_wait_ms"] / 1e3
                         with outer._cond:
                             while True:
+                                # max_updates=0 = no limit (thrift:36-38);
+                                # the db layer bounds it by the byte cap
                                 ups = db.get_updates(req["seq_no"],
-                                                     req["max_updates"] or 50,
+                                                     req["max_updates"],
                                                      observer=obs)
                                 if ups:
                                     break

@@ -121,7 +121,8 @@ class _FakeDb:
         self.log = []  # (seq, ts, rep)
 
     def get_updates(self, since, max_updates=50, observer=False):
-        return [u for u in self.log if u[0] > since][:max_updates]
+        ups = [u for u in self.log if u[0] > since]
+        return ups[:max_updates] if max_updates else ups
 
 
 def test_server_client_end_to_end():
