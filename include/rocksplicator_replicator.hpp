@@ -135,13 +135,16 @@ class GpuReplicator {
     int rc = gra_write_leader(rs->db, rep, len, &seq);
     if (rc != GRA_OK) throw std::runtime_error(gra_last_error());
     if (mode == 1 || mode == 2) {
-      int timeout = consecutive_misses_.load() >= degrade_after_misses
+      /* degradation state is PER shard, as the reference keeps it per
+       * ReplicatedDB (replicated_db.cpp:236-273) — one shard's healthy
+       * acks must not reset another's degradation */
+      int timeout = rs->consecutive_misses.load() >= degrade_after_misses
                         ? degraded_timeout_ms
                         : ack_timeout_ms;
       if (gra_wait_ack(rs->db, seq, mode == 2 ? 1 : 0, timeout) == GRA_OK)
-        consecutive_misses_.store(0);
+        rs->consecutive_misses.store(0);
       else
-        consecutive_misses_.fetch_add(1);
+        rs->consecutive_misses.fetch_add(1);
     }
     return seq;
   }
@@ -167,6 +170,7 @@ class GpuReplicator {
     UpstreamFn upstream;
     std::thread thread;
     std::atomic<bool> stop{false};
+    std::atomic<int> consecutive_misses{0}; /* write-degradation, per db */
   };
 
   /* The pullFromUpstream control flow (replicated_db.cpp:314-433): request
@@ -209,7 +213,6 @@ class GpuReplicator {
   std::mutex mu_;
   std::map<std::string, std::shared_ptr<Shard>> dbs_;
   uint32_t next_shard_ = 0;
-  std::atomic<int> consecutive_misses_{0};
 };
 
 } /* namespace gra */
