@@ -201,6 +201,17 @@ void gra_wb_single_delete(GraBatch *b, const void *k, size_t klen);
 void gra_wb_merge(GraBatch *b, const void *k, size_t klen, const void *v, size_t vlen);
 void gra_wb_delete_range(GraBatch *b, const void *bk, size_t bklen,
                          const void *ek, size_t eklen);
+/* CF-prefixed record variants (varint cf id before the slices; cf 0 uses
+ * the plain forms, as rocksdb does). */
+void gra_wb_cf_put(GraBatch *b, uint32_t cf, const void *k, size_t klen,
+                   const void *v, size_t vlen);
+void gra_wb_cf_delete(GraBatch *b, uint32_t cf, const void *k, size_t klen);
+void gra_wb_cf_single_delete(GraBatch *b, uint32_t cf, const void *k,
+                             size_t klen);
+void gra_wb_cf_merge(GraBatch *b, uint32_t cf, const void *k, size_t klen,
+                     const void *v, size_t vlen);
+void gra_wb_cf_delete_range(GraBatch *b, uint32_t cf, const void *bk,
+                            size_t bklen, const void *ek, size_t eklen);
 void gra_wb_put_log_data(GraBatch *b, const void *blob, size_t blen);
 void gra_wb_set_seq(GraBatch *b, uint64_t seq);
 uint32_t gra_wb_count(const GraBatch *b);

@@ -71,6 +71,46 @@ void gra_wb_delete_range(GraBatch *b, const void *bk, size_t bkl, const void *ek
   b->slice(ek, ekl);
   b->set_count(b->count + 1);
 }
+/* CF-prefixed variants (tags 0x04-0x06/0x08/0x0E: varint32 cf id before
+ * the slices). cf_id 0 is the default family — callers use the plain
+ * forms for it, as rocksdb does. */
+void gra_wb_cf_put(GraBatch *b, uint32_t cf, const void *k, size_t kl,
+                   const void *v, size_t vl) {
+  b->buf.push_back(wb::kCfValue);
+  b->varint(cf);
+  b->slice(k, kl);
+  b->slice(v, vl);
+  b->set_count(b->count + 1);
+}
+void gra_wb_cf_delete(GraBatch *b, uint32_t cf, const void *k, size_t kl) {
+  b->buf.push_back(wb::kCfDeletion);
+  b->varint(cf);
+  b->slice(k, kl);
+  b->set_count(b->count + 1);
+}
+void gra_wb_cf_single_delete(GraBatch *b, uint32_t cf, const void *k,
+                             size_t kl) {
+  b->buf.push_back(wb::kCfSingleDeletion);
+  b->varint(cf);
+  b->slice(k, kl);
+  b->set_count(b->count + 1);
+}
+void gra_wb_cf_merge(GraBatch *b, uint32_t cf, const void *k, size_t kl,
+                     const void *v, size_t vl) {
+  b->buf.push_back(wb::kCfMerge);
+  b->varint(cf);
+  b->slice(k, kl);
+  b->slice(v, vl);
+  b->set_count(b->count + 1);
+}
+void gra_wb_cf_delete_range(GraBatch *b, uint32_t cf, const void *bk,
+                            size_t bkl, const void *ek, size_t ekl) {
+  b->buf.push_back(wb::kCfRangeDeletion);
+  b->varint(cf);
+  b->slice(bk, bkl);
+  b->slice(ek, ekl);
+  b->set_count(b->count + 1);
+}
 void gra_wb_put_log_data(GraBatch *b, const void *blob, size_t bl) {
   b->buf.push_back(wb::kLogData);
   b->slice(blob, bl); /* consumes no count */

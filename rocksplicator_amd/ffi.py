@@ -156,6 +156,11 @@ def load():
     for f in ("gra_wb_delete", "gra_wb_single_delete"):
         getattr(lib, f).argtypes = [C.c_void_p, C.c_char_p, C.c_size_t]
     lib.gra_wb_delete_range.argtypes = [C.c_void_p, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    lib.gra_wb_cf_put.argtypes = [C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    lib.gra_wb_cf_delete.argtypes = [C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t]
+    lib.gra_wb_cf_single_delete.argtypes = [C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t]
+    lib.gra_wb_cf_merge.argtypes = [C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    lib.gra_wb_cf_delete_range.argtypes = [C.c_void_p, C.c_uint32, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
     lib.gra_wb_put_log_data.argtypes = [C.c_void_p, C.c_char_p, C.c_size_t]
     lib.gra_wb_set_seq.argtypes = [C.c_void_p, C.c_uint64]
     lib.gra_wb_count.argtypes = [C.c_void_p]
@@ -214,6 +219,26 @@ class Batch:
 
     def log_data(self, blob):
         self.lib.gra_wb_put_log_data(self.h, blob, len(blob))
+        return self
+
+    def cf_put(self, cf, k, v):
+        self.lib.gra_wb_cf_put(self.h, cf, k, len(k), v, len(v))
+        return self
+
+    def cf_delete(self, cf, k):
+        self.lib.gra_wb_cf_delete(self.h, cf, k, len(k))
+        return self
+
+    def cf_single_delete(self, cf, k):
+        self.lib.gra_wb_cf_single_delete(self.h, cf, k, len(k))
+        return self
+
+    def cf_merge(self, cf, k, v):
+        self.lib.gra_wb_cf_merge(self.h, cf, k, len(k), v, len(v))
+        return self
+
+    def cf_delete_range(self, cf, bk, ek):
+        self.lib.gra_wb_cf_delete_range(self.h, cf, bk, len(bk), ek, len(ek))
         return self
 
     def set_seq(self, seq):

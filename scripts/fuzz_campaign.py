@@ -53,33 +53,33 @@ def rand_ops(rng, keys, n):
 
 
 def build(ops, seq=0):
-    """Returns (rep, product_rep_or_None): the product builder has no CF
-    API (the reference's CF batches arrive over the wire, not from this
-    builder), so batches containing cf ops are built by pywb alone and
-    skip the builder-equality check."""
-    has_cf = any(kind.startswith("cf_") for kind, _a, _b in ops)
+    """Both builders (pywb + product gra_wb_*) for every op kind,
+    byte-compared."""
     pb = PyBatch(seq=seq)
-    gb = None if has_cf else ra.Batch().set_seq(seq)
+    gb = ra.Batch().set_seq(seq)
     for kind, a, b in ops:
         if kind in ("put", "merge"):
             getattr(pb, kind)(a, b)
-            if gb: getattr(gb, kind)(a, b)
+            getattr(gb, kind)(a, b)
         elif kind == "delete_range":
             pb.delete_range(a, b)
-            if gb: gb.delete_range(a, b)
+            gb.delete_range(a, b)
         elif kind == "log_data":
             pb.log_data(a)
-            if gb: gb.log_data(a)
+            gb.log_data(a)
         elif kind in ("cf_put", "cf_merge"):
             getattr(pb, kind)(a[0], a[1], b)
+            getattr(gb, kind)(a[0], a[1], b)
         elif kind in ("cf_delete", "cf_single_delete"):
             getattr(pb, kind)(a[0], a[1])
+            getattr(gb, kind)(a[0], a[1])
         elif kind == "cf_delete_range":
             pb.cf_delete_range(a[0], a[1], a[2])
+            gb.cf_delete_range(a[0], a[1], a[2])
         else:
             getattr(pb, kind)(a)
-            if gb: getattr(gb, kind)(a)
-    return pb.data(), (gb.data() if gb else None)
+            getattr(gb, kind)(a)
+    return pb.data(), gb.data()
 
 
 def main():
@@ -101,9 +101,7 @@ def main():
         for _batch in range(rng.randrange(1, 6)):
             ops = rand_ops(rng, keys, rng.randrange(1, 8))
             rep_py, rep_c = build(ops)
-            if rep_c is not None:
-                assert rep_py == rep_c, ("builder divergence", case, ops)
-            rep_c = rep_py
+            assert rep_py == rep_c, ("builder divergence", case, ops)
             if rng.random() < 0.25:  # corrupt sometimes
                 rep = bytearray(rep_c)
                 if rep and rng.random() < 0.5:

@@ -182,7 +182,12 @@ def test_product_builder_against_hand_kats():
         "put_0x01": lambda b: b.put(b"k", b"v"),
         "merge_0x02": lambda b: b.merge(b"m", b"x"),
         "logdata_0x03": lambda b: b.log_data(b"LOG"),
+        "cf_delete_0x04": lambda b: b.cf_delete(5, b"k"),
+        "cf_put_0x05": lambda b: b.cf_put(1, b"k", b"v"),
+        "cf_merge_0x06_varint_cf": lambda b: b.cf_merge(200, b"k", b"v"),
         "single_delete_0x07": lambda b: b.single_delete(b"k"),
+        "cf_single_delete_0x08": lambda b: b.cf_single_delete(3, b"k"),
+        "cf_range_delete_0x0E": lambda b: b.cf_delete_range(2, b"a", b"b"),
         "range_delete_0x0F": lambda b: b.delete_range(b"a", b"b"),
         "put_val127": lambda b: b.put(b"k", b"A" * 127),
         "put_val128": lambda b: b.put(b"k", b"A" * 128),
