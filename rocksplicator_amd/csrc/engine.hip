@@ -2325,7 +2325,11 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
   uint32_t qtab_size = 64;
   while (qtab_size < 2 * nq) qtab_size <<= 1;
   uint32_t cand_cap = 4 * nq + 1024, tomb_cap = 4096;
-  bool hashjoin = views.size() <= 65535;
+  static const bool force_scan = [] { /* debug: bypass the hash join */
+    const char *v = getenv("GRA_MG_FORCE_SCAN");
+    return v && v[0] == '1';
+  }();
+  bool hashjoin = views.size() <= 65535 && !force_scan;
   if ((mixed && !grow(&mg.d_extra, &mg.extra_cap, nq * sizeof(MgExtra))) ||
       (hashjoin &&
        (!grow(&mg.d_qtab, &mg.qtab_cap, qtab_size * 8) ||
