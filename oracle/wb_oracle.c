@@ -496,6 +496,18 @@ int orc_get(const OrcStore *s, uint32_t shard, const void *key_, size_t klen,
     break; /* VALUE / DELETION / SINGLE_DELETION all stop the walk */
   }
   if (!have_base && nops == 0) return 1; /* not found / deleted */
+  if (nops == 0) {
+    /* No merge operands above the base: rocksdb returns the Put's value
+     * VERBATIM — the merge operator (FullMerge) only runs when merge
+     * records are newer than the base (db/memtable.cc semantics). The
+     * earlier restatement folded plain Puts to 8 bytes under u64add,
+     * which the round-2 GPU fuzz soak exposed as a deviation. */
+    Entry *be = &t->entries[base_entry - 1];
+    if (cap < be->val_len) return 2;
+    memcpy(buf, t->arena + be->val_off, be->val_len);
+    if (vlen) *vlen = be->val_len;
+    return 0;
+  }
   /* fold operator, oldest -> newest */
   if (s->merge_op == ORC_MERGE_U64ADD) {
     uint64_t acc = 0;

@@ -487,7 +487,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
                            const uint8_t *__restrict__ keybuf, uint32_t nq,
                            uint8_t *__restrict__ valbuf, uint32_t val_stride,
                            GraGetResult *__restrict__ out,
-                           MgExtra *__restrict__ extra) {
+                           MgExtra *__restrict__ extra, int use_kpref) {
   __shared__ uint64_t sh_term_seq[256];
   __shared__ uint64_t sh_term_ref[256]; /* (run<<32)|entry, ~0 = none */
   __shared__ uint64_t sh_merge_seq[256];
@@ -505,7 +505,7 @@ __global__ void k_multiget(const uint8_t *__restrict__ store,
     for (uint32_t i = threadIdx.x; i < rv.n_entries; i += blockDim.x) {
       wb::RecHdr h = hdrs[i];
       if (h.type != wb::kRangeDeletion &&
-          (h.key_len != klen || h.kpref != qpref))
+          (h.key_len != klen || (use_kpref && h.kpref != qpref)))
         continue; /* header-only reject: no payload touch */
       uint32_t rel = h.kv_off - rv.pay_rel_base;
       if (h.type == wb::kRangeDeletion) {
@@ -2457,10 +2457,14 @@ int gra_multiget(GraDb *db, uint32_t nq, const GraKeyRef *keys,
                            hipMemcpyDeviceToHost, e->stream) != hipSuccess)
           break;
       } else {
+        static const int use_kpref = [] { /* debug: bypass the filter */
+          const char *v = getenv("GRA_MG_NO_KPREF");
+          return (v && v[0] == '1') ? 0 : 1;
+        }();
         hipLaunchKernelGGL(k_multiget, dim3(nq), dim3(256), 0, e->stream,
                            e->d_store, d_runs, (uint32_t)views.size(), d_keys,
                            d_keybuf, nq, d_valbuf, val_stride, d_out,
-                           mixed ? (MgExtra *)mg.d_extra : nullptr);
+                           mixed ? (MgExtra *)mg.d_extra : nullptr, use_kpref);
         if (hipGetLastError() != hipSuccess) break;
       }
       if (hipMemcpyAsync(out, d_out, nq * sizeof(GraGetResult),

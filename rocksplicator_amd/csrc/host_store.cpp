@@ -61,6 +61,13 @@ int run_get(const std::vector<std::shared_ptr<Run>> &runs, const void *key_,
   }
 
   if (!have_base && ops.empty()) return 1;
+  if (ops.empty()) {
+    /* no operands above the base: the Put's value verbatim (rocksdb
+     * FullMerge only runs when merge records are newer — mirrored in the
+     * oracle and the device path) */
+    out->assign((const char *)base, base_len);
+    return 0;
+  }
   if (merge_op == 1 /* u64add */) {
     uint64_t acc = 0;
     if (have_base) memcpy(&acc, base, base_len < 8 ? base_len : 8);

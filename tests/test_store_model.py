@@ -86,6 +86,9 @@ class ModelStore:
             break  # put/delete/single_delete all stop the walk
         if base is None and not operands:
             return None
+        if not operands:
+            return base  # a plain Put reads back verbatim (rocksdb: the
+            # merge operator only runs when operands are newer)
         if self.merge_op == oracle_ffi.Store.MERGE_U64ADD:
             acc = sum(int.from_bytes(v[:8], "little")
                       for v in ([base] if base is not None else []) + operands)
