@@ -21,13 +21,17 @@ import rocksplicator_amd as ra  # noqa: E402
 from pywb import PyBatch  # noqa: E402
 
 
-def main(seconds=300, seed=0x50AC):
+def main(seconds=300, seed=0x50AC, mode="plain"):
+    """mode: plain | drain (drain_host runs, arena-backed reads) |
+    concat (merge_op 0)."""
     rng = random.Random(seed)
     olib = oracle_ffi.load()
     nshards = 12
-    e = ra.Engine(nshards=nshards, merge_op=1)
+    merge_op = 0 if mode == "concat" else 1
+    e = ra.Engine(nshards=nshards, merge_op=merge_op,
+                  drain_host=1 if mode == "drain" else 0)
     dbs = [e.open(s) for s in range(nshards)]
-    ost = oracle_ffi.Store(olib, nshards, merge_op=1)
+    ost = oracle_ffi.Store(olib, nshards, merge_op=merge_op)
     keys = [f"key{i:04d}".encode() for i in range(300)]
     one = (1).to_bytes(8, "little")
     deadline = time.monotonic() + seconds
@@ -95,8 +99,8 @@ def main(seconds=300, seed=0x50AC):
                    for cf in (1, 2, 3)]
         for k, v in zip(probes, dbs[s].multiget(probes)):
             assert v == ost.get(s, k), (rounds, s, k)
-    print(f"gpu fuzz soak clean: {rounds} rounds, {batches} batches "
-          f"({mutations} mutated), {checks} shard checksum checks, "
+    print(f"gpu fuzz soak clean [{mode}]: {rounds} rounds, {batches} "
+          f"batches ({mutations} mutated), {checks} shard checksum checks, "
           f"seed {seed:#x}")
     e.close()
     return 0
@@ -104,4 +108,5 @@ def main(seconds=300, seed=0x50AC):
 
 if __name__ == "__main__":
     sys.exit(main(int(sys.argv[1]) if len(sys.argv) > 1 else 300,
-                  int(sys.argv[2], 0) if len(sys.argv) > 2 else 0x50AC))
+                  int(sys.argv[2], 0) if len(sys.argv) > 2 else 0x50AC,
+                  sys.argv[3] if len(sys.argv) > 3 else "plain"))
