@@ -38,6 +38,16 @@ def main(seconds=300, seed=0x50AC, mode="plain"):
     batches = mutations = checks = rounds = 0
     while time.monotonic() < deadline:
         rounds += 1
+        if mode == "drain" and rounds % 20 == 0:
+            # drain mode accumulates pinned host arenas per tick (runs own
+            # them); bound the footprint by recycling the engine — which
+            # also exercises the restart story mid-soak
+            for db in dbs:
+                db.close()
+            e.close()
+            ost = oracle_ffi.Store(olib, nshards, merge_op=merge_op)
+            e = ra.Engine(nshards=nshards, merge_op=merge_op, drain_host=1)
+            dbs = [e.open(s) for s in range(nshards)]
         for _ in range(rng.randrange(100, 400)):
             s = rng.randrange(nshards)
             b = PyBatch()
