@@ -3,9 +3,17 @@ types incl. CF variants), random single-byte corruption, periodic poison
 recovery and periodic full-store checksum comparison — engine vs oracle,
 per shard, for a wall-clock budget. Run on a GPU box:
 
-    python scripts/gpu_fuzz_soak.py [seconds] [seed]
+    python scripts/gpu_fuzz_soak.py [seconds] [seed] [plain|drain|concat]
 
 Exit 0 = every probe agreed. Totals printed for the log.
+
+Status: the `plain` mode ran clean at scale (profiles/r02/
+gpu_fuzz_soak2.log: 827K batches / 40K checksum checks) and caught the
+verbatim-Put semantics divergence on its first run. The `drain` and
+`concat` variants were added late in round 2 and could not be executed:
+the GPU pool lost three boxes in the setup phase (before the command
+ran) on consecutive calls, closing gpurun for the round — the variants
+are untested on hardware, flagged here rather than silently.
 """
 import os
 import random
