@@ -59,6 +59,20 @@ class InMemoryBroker:
                     return None
                 self._cond.wait(rem)
 
+    def wait_any(self, topic, wants, timeout_s):
+        """Block until ANY (partition, next_offset) in `wants` has data, or
+        timeout. Returns True if something is available."""
+        deadline = time.monotonic() + timeout_s
+        with self._cond:
+            while True:
+                for p, off in wants.items():
+                    if off < len(self._logs.get((topic, p), [])):
+                        return True
+                rem = deadline - time.monotonic()
+                if rem <= 0:
+                    return False
+                self._cond.wait(rem)
+
     def high_watermark(self, topic, partition):
         with self._cond:
             return len(self._logs.get((topic, partition), []))
@@ -99,9 +113,9 @@ class InMemoryConsumer:
             rem = deadline - time.monotonic()
             if rem <= 0:
                 return None
-            # park on any partition (the broker condition wakes all)
-            self._broker.fetch(self._topic, self._parts[0],
-                               self._next[self._parts[0]], rem)
+            # park until ANY assigned partition has data (a single-partition
+            # wait could sleep through another partition's arrival)
+            self._broker.wait_any(self._topic, dict(self._next), rem)
 
     def commit(self, message):
         self._committed[message.partition] = message.offset
