@@ -210,3 +210,33 @@ def test_large_binary_and_field_long_form():
 
     tw.read_struct_fields(tw.Reader(bytes(w.out)), f)
     assert got == {"b": big, "x": 7}
+
+
+def test_transport_equivalence_with_framework_wire():
+    """The thrift framing and the framework's length-prefixed framing must
+    deliver IDENTICAL update streams from the same db (transport choice
+    cannot change replication content)."""
+    from rocksplicator_amd import wire
+
+    db = _FakeDb()
+    db.log = [(i + 1, 100 + i, bytes([i % 251]) * (i % 37 + 1))
+              for i in range(120)]
+    tsrv = tw.ThriftUpdateServer()
+    wsrv = wire.UpdateServer()
+    try:
+        tsrv.register("d", db)
+        wsrv.register("d", db)
+        tcli = tw.ThriftRemoteUpstream("127.0.0.1", tsrv.port, "d")
+        wcli = wire.RemoteUpstream("127.0.0.1", wsrv.port, "d")
+        try:
+            for since in (0, 1, 57, 119, 120, 500):
+                for mx in (1, 14, 50, 0):
+                    a = tcli.get_updates(since, mx)
+                    b = wcli.get_updates(since, mx)
+                    assert a == b, (since, mx)
+        finally:
+            tcli.close()
+            wcli.close()
+    finally:
+        tsrv.close()
+        wsrv.close()
