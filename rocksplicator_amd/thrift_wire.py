@@ -486,7 +486,10 @@ class ThriftUpdateServer:
                         reply = encode_reply_replicate(mseq, ups,
                                                        role=ROLE_LEADER)
                         self.request.sendall(frame(reply, seq_id))
-                except (ConnectionError, OSError, ValueError):
+                except (ConnectionError, OSError, ValueError, IndexError,
+                        struct.error):
+                    # malformed frame or peer gone: drop the connection
+                    # (the reference's server closes the channel likewise)
                     pass
 
         class Srv(socketserver.ThreadingTCPServer):
