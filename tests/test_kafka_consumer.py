@@ -116,3 +116,23 @@ def test_rdkafka_consumer_absent_is_loud():
         raise AssertionError("expected RuntimeError (no confluent_kafka)")
     except RuntimeError as e:
         assert "confluent_kafka" in str(e)
+
+
+def test_watcher_commit_cadence_and_final_commit():
+    """The watcher commits broker offsets every `commit_every` applies and
+    once more on stop — a restart from the BROKER's committed position
+    then redelivers at most `commit_every-1` messages (all deduped)."""
+    b = InMemoryBroker()
+    for i in range(10):
+        b.produce("t", 0, f"m{i}".encode())
+    ing = FakeIngestor([0])
+    c = InMemoryConsumer(b, "t", [0])
+    w = KafkaWatcher(c, ing, commit_every=4, poll_ms=50)
+    w.start()
+    deadline = time.monotonic() + 5
+    while len(ing.applied) < 10 and time.monotonic() < deadline:
+        time.sleep(0.02)
+    w.stop()
+    # broker-side committed offset reflects the final commit on stop
+    assert c.committed()[0] == 9
+    assert w.applied == 10 and w.duplicates == 0
